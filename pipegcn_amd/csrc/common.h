@@ -1,0 +1,70 @@
+// Common declarations shared between the host graph core and the HIP kernel
+// launchers of pipegcn_amd._C.
+//
+// The native layer replaces the reference's external stacks (DGL C++/CUDA SpMM,
+// METIS partitioning, Gloo staging; see /root/reference SURVEY §2.2) with
+// MI355X-native C++/HIP code compiled for gfx950 only.
+#pragma once
+
+#include <torch/extension.h>
+
+#include <cstdint>
+#include <vector>
+
+// ---------------------------------------------------------------------------
+// Host-side graph core (graph_core.cpp)
+// ---------------------------------------------------------------------------
+
+// Build a CSR adjacency over destination rows from a COO edge list.
+//   u, v : int64/int32 CPU tensors of equal length E (edge u -> v)
+//   num_rows : number of destination nodes (rows)
+// Returns (indptr int64[num_rows+1], indices int32[E]) where
+// indices[indptr[r]:indptr[r+1]] are the sources of row r (unsorted within row).
+std::vector<torch::Tensor> build_csr(torch::Tensor u, torch::Tensor v,
+                                     int64_t num_rows);
+
+// Multi-way graph partitioner (replaces METIS; BFS-grown balanced seeds +
+// boundary refinement). Operates on an undirectedized CSR of the full graph.
+//   indptr int64[N+1], indices int32[E] (CSR over rows = nodes)
+//   nparts : number of partitions
+//   objective : 0 = edge cut, 1 = communication volume ("vol")
+//   balance_slack : allowed imbalance, e.g. 0.05
+// Returns int32[N] partition assignment.
+torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
+                                  int64_t nparts, int64_t objective,
+                                  double balance_slack, int64_t n_refine_passes,
+                                  int64_t seed);
+
+// CPU CSR SpMM: out[r, :] = scale[r] * sum_{c in row r} feat[indices[c], :]
+// scale may be an undefined tensor (no scaling). feat float32 [num_src, F].
+torch::Tensor spmm_cpu(torch::Tensor indptr, torch::Tensor indices,
+                       torch::Tensor feat, torch::Tensor scale,
+                       int64_t num_rows);
+
+// ---------------------------------------------------------------------------
+// HIP kernel launchers (hip/kernels.hip) — gfx950 only.
+// ---------------------------------------------------------------------------
+
+// out[r,:] = scale[r] * sum_{e in [indptr[r], indptr[r+1])} feat[indices[e],:]
+// All tensors on device. feat fp32 [num_src, F], out fp32 [num_rows, F].
+void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
+                  torch::Tensor feat, torch::Tensor scale, torch::Tensor out);
+
+// out[i,:] = src[idx[i],:]
+void gather_rows_hip(torch::Tensor src, torch::Tensor idx, torch::Tensor out);
+
+// dst[idx[i],:] += src[i,:]   (idx entries unique — no atomics needed? they
+// are unique for the boundary scatter; kernel uses one row per wave so
+// duplicates would race — documented contract: unique indices.)
+void scatter_add_rows_hip(torch::Tensor dst, torch::Tensor idx,
+                          torch::Tensor src);
+
+// avg = momentum * avg + (1 - momentum) * x
+void ema_update_hip(torch::Tensor avg, torch::Tensor x, double momentum);
+
+// Fused dual GEMM for the GraphSAGE layer epilogue:
+//   out[M,N] = x1[M,K] @ w1t[K,N] + x2[M,K] @ w2t[K,N] + b[N]
+// fp32, MFMA (v_mfma_f32_16x16x4_f32). Weights pre-transposed to [K,N].
+void sage_dual_gemm_hip(torch::Tensor x1, torch::Tensor x2, torch::Tensor w1t,
+                        torch::Tensor w2t, torch::Tensor bias,
+                        torch::Tensor out);

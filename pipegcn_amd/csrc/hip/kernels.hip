@@ -1,0 +1,299 @@
+// Hand-written HIP/CDNA4 (gfx950) kernels for the pipegcn_amd hot path.
+//
+// These replace the reference's DGL CUDA gSpMM (copy_src+sum, invoked at
+// /root/reference/module/layer.py:47-49), its index gather/scatter
+// (/root/reference/helper/feature_buffer.py:176,215-217) and the EMA
+// smoothing correction (/root/reference/helper/feature_buffer.py:189-191).
+//
+// Design (MI355X-first, see /opt/skills/guides/cdna_hip_programming.md):
+//  - wavefront = 64 lanes; feature dim is mapped across lanes with float4/2/1
+//    vector loads so each wave issues 1 KiB coalesced reads of a neighbor row.
+//  - SpMM is a bandwidth/gather-bound op: one wave owns one (dst-row, feature
+//    chunk); the edge loop is unrolled 4x to keep 4 gathers in flight per
+//    lane, and occupancy stays high (tiny VGPR footprint) so TLP hides HBM
+//    latency. The degree-divide epilogue is fused (scale argument).
+//  - The backward (transpose) SpMM is the same kernel run over the CSC of the
+//    halo graph, built once at setup — no atomics anywhere on the hot path.
+
+#include "../common.h"
+
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+#include <hip/hip_runtime.h>
+
+#define HIP_CHECK(expr)                                              \
+  do {                                                               \
+    hipError_t _e = (expr);                                          \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e)); \
+  } while (0)
+
+namespace {
+
+constexpr int kWave = 64;
+
+inline hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+// ---------------------------------------------------------------------------
+// CSR SpMM: out[r, fchunk] = scale[r] * sum_e feat[indices[e], fchunk]
+// One wave per (row, chunk) pair, grid-stride. VEC in {4, 2, 1}.
+// ---------------------------------------------------------------------------
+
+template <int VEC>
+__global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
+                                const int32_t* __restrict__ indices,
+                                const float* __restrict__ feat,
+                                const float* __restrict__ scale,
+                                float* __restrict__ out, int64_t num_rows,
+                                int64_t F, int64_t nchunks) {
+  const int64_t wave_global =
+      (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / kWave;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t nwaves =
+      (static_cast<int64_t>(gridDim.x) * blockDim.x) / kWave;
+  const int64_t npairs = num_rows * nchunks;
+
+  for (int64_t pair = wave_global; pair < npairs; pair += nwaves) {
+    const int64_t r = pair / nchunks;
+    const int64_t chunk = pair % nchunks;
+    const int64_t f0 = chunk * (kWave * VEC) + lane * VEC;
+    if (f0 >= F) continue;
+    const bool full = (f0 + VEC <= F);
+
+    float acc[VEC];
+#pragma unroll
+    for (int k = 0; k < VEC; ++k) acc[k] = 0.f;
+
+    const int64_t e_begin = indptr[r];
+    const int64_t e_end = indptr[r + 1];
+    int64_t e = e_begin;
+
+    if (full) {
+      // main path: unroll 4 edges so 4 gathers are in flight per lane
+      for (; e + 4 <= e_end; e += 4) {
+        const int64_t u0 = indices[e + 0];
+        const int64_t u1 = indices[e + 1];
+        const int64_t u2 = indices[e + 2];
+        const int64_t u3 = indices[e + 3];
+        float v0[VEC], v1[VEC], v2[VEC], v3[VEC];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) v0[k] = feat[u0 * F + f0 + k];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) v1[k] = feat[u1 * F + f0 + k];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) v2[k] = feat[u2 * F + f0 + k];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) v3[k] = feat[u3 * F + f0 + k];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) acc[k] += v0[k] + v1[k] + v2[k] + v3[k];
+      }
+      for (; e < e_end; ++e) {
+        const int64_t u = indices[e];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) acc[k] += feat[u * F + f0 + k];
+      }
+      const float s = scale ? scale[r] : 1.f;
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) out[r * F + f0 + k] = acc[k] * s;
+    } else {
+      // ragged tail chunk: scalar guarded
+      for (; e < e_end; ++e) {
+        const int64_t u = indices[e];
+        for (int k = 0; k < VEC && f0 + k < F; ++k)
+          acc[k] += feat[u * F + f0 + k];
+      }
+      const float s = scale ? scale[r] : 1.f;
+      for (int k = 0; k < VEC && f0 + k < F; ++k)
+        out[r * F + f0 + k] = acc[k] * s;
+    }
+  }
+}
+
+template <int VEC>
+void launch_spmm(const int64_t* indptr, const int32_t* indices,
+                 const float* feat, const float* scale, float* out,
+                 int64_t num_rows, int64_t F, hipStream_t stream) {
+  const int64_t nchunks = (F + kWave * VEC - 1) / (kWave * VEC);
+  const int threads = 256;  // 4 waves
+  const int64_t npairs = num_rows * nchunks;
+  int64_t blocks = (npairs * kWave + threads - 1) / threads;
+  // MI355X: 256 CUs; >= 2048 workgroups fills the chip with headroom.
+  blocks = std::min<int64_t>(blocks, 8 * 65536);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(HIP_KERNEL_NAME(spmm_csr_kernel<VEC>), dim3(blocks),
+                     dim3(threads), 0, stream, indptr, indices, feat, scale,
+                     out, num_rows, F, nchunks);
+  HIP_CHECK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
+// Row gather / scatter-add. One wave per (row, chunk).
+// ---------------------------------------------------------------------------
+
+template <int VEC, bool SCATTER_ADD>
+__global__ void rowcopy_kernel(const float* __restrict__ src,
+                               const int64_t* __restrict__ idx,
+                               float* __restrict__ dst, int64_t nrows,
+                               int64_t F, int64_t nchunks) {
+  const int64_t wave_global =
+      (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / kWave;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t nwaves =
+      (static_cast<int64_t>(gridDim.x) * blockDim.x) / kWave;
+  const int64_t npairs = nrows * nchunks;
+  for (int64_t pair = wave_global; pair < npairs; pair += nwaves) {
+    const int64_t i = pair / nchunks;
+    const int64_t f0 = pair % nchunks * (kWave * VEC) + lane * VEC;
+    if (f0 >= F) continue;
+    const int64_t j = idx[i];
+    if (f0 + VEC <= F) {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        if (SCATTER_ADD)
+          dst[j * F + f0 + k] += src[i * F + f0 + k];
+        else
+          dst[i * F + f0 + k] = src[j * F + f0 + k];
+      }
+    } else {
+      for (int k = 0; f0 + k < F; ++k) {
+        if (SCATTER_ADD)
+          dst[j * F + f0 + k] += src[i * F + f0 + k];
+        else
+          dst[i * F + f0 + k] = src[j * F + f0 + k];
+      }
+    }
+  }
+}
+
+template <int VEC, bool SCATTER_ADD>
+void launch_rowcopy(const float* src, const int64_t* idx, float* dst,
+                    int64_t nrows, int64_t F, hipStream_t stream) {
+  const int64_t nchunks = (F + kWave * VEC - 1) / (kWave * VEC);
+  const int threads = 256;
+  int64_t blocks = (nrows * nchunks * kWave + threads - 1) / threads;
+  blocks = std::min<int64_t>(blocks, 8 * 65536);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(HIP_KERNEL_NAME(rowcopy_kernel<VEC, SCATTER_ADD>),
+                     dim3(blocks), dim3(threads), 0, stream, src, idx, dst,
+                     nrows, F, nchunks);
+  HIP_CHECK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
+// EMA: avg = m * avg + (1 - m) * x   (flat elementwise, float4)
+// ---------------------------------------------------------------------------
+
+__global__ void ema_kernel(float* __restrict__ avg,
+                           const float* __restrict__ x, float m, int64_t n) {
+  const int64_t i4 = (static_cast<int64_t>(blockIdx.x) * blockDim.x +
+                      threadIdx.x) * 4;
+  if (i4 + 4 <= n) {
+    float4 a = *reinterpret_cast<float4*>(avg + i4);
+    const float4 b = *reinterpret_cast<const float4*>(x + i4);
+    a.x = m * a.x + (1.f - m) * b.x;
+    a.y = m * a.y + (1.f - m) * b.y;
+    a.z = m * a.z + (1.f - m) * b.z;
+    a.w = m * a.w + (1.f - m) * b.w;
+    *reinterpret_cast<float4*>(avg + i4) = a;
+  } else {
+    for (int64_t i = i4; i < n; ++i) avg[i] = m * avg[i] + (1.f - m) * x[i];
+  }
+}
+
+int pick_vec(int64_t F) {
+  if (F % 4 == 0) return 4;
+  if (F % 2 == 0) return 2;
+  return 1;
+}
+
+}  // namespace
+
+void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
+                  torch::Tensor feat, torch::Tensor scale, torch::Tensor out) {
+  TORCH_CHECK(feat.is_cuda() && out.is_cuda(), "spmm_csr_hip: device tensors");
+  TORCH_CHECK(feat.scalar_type() == torch::kFloat, "fp32 only");
+  TORCH_CHECK(feat.is_contiguous() && out.is_contiguous());
+  TORCH_CHECK(indptr.scalar_type() == torch::kLong &&
+              indices.scalar_type() == torch::kInt);
+  const int64_t num_rows = out.size(0);
+  const int64_t F = feat.size(1);
+  TORCH_CHECK(out.size(1) == F && indptr.numel() == num_rows + 1);
+  const float* sp = nullptr;
+  if (scale.defined() && scale.numel() > 0) {
+    TORCH_CHECK(scale.is_contiguous() && scale.numel() == num_rows);
+    sp = scale.data_ptr<float>();
+  }
+  auto stream = current_stream();
+  const int vec = pick_vec(F);
+  if (vec == 4)
+    launch_spmm<4>(indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
+                   feat.data_ptr<float>(), sp, out.data_ptr<float>(), num_rows,
+                   F, stream);
+  else if (vec == 2)
+    launch_spmm<2>(indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
+                   feat.data_ptr<float>(), sp, out.data_ptr<float>(), num_rows,
+                   F, stream);
+  else
+    launch_spmm<1>(indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
+                   feat.data_ptr<float>(), sp, out.data_ptr<float>(), num_rows,
+                   F, stream);
+}
+
+void gather_rows_hip(torch::Tensor src, torch::Tensor idx, torch::Tensor out) {
+  TORCH_CHECK(src.is_cuda() && idx.is_cuda() && out.is_cuda());
+  TORCH_CHECK(src.scalar_type() == torch::kFloat &&
+              idx.scalar_type() == torch::kLong);
+  TORCH_CHECK(src.is_contiguous() && out.is_contiguous());
+  const int64_t F = src.size(1);
+  const int64_t n = idx.numel();
+  TORCH_CHECK(out.size(0) == n && out.size(1) == F);
+  auto stream = current_stream();
+  const int vec = pick_vec(F);
+  if (vec == 4)
+    launch_rowcopy<4, false>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                             out.data_ptr<float>(), n, F, stream);
+  else if (vec == 2)
+    launch_rowcopy<2, false>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                             out.data_ptr<float>(), n, F, stream);
+  else
+    launch_rowcopy<1, false>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                             out.data_ptr<float>(), n, F, stream);
+}
+
+void scatter_add_rows_hip(torch::Tensor dst, torch::Tensor idx,
+                          torch::Tensor src) {
+  // Contract: idx entries are unique (one boundary peer at a time) — rows
+  // are written by exactly one wave, so no atomics are needed.
+  TORCH_CHECK(dst.is_cuda() && idx.is_cuda() && src.is_cuda());
+  TORCH_CHECK(dst.scalar_type() == torch::kFloat &&
+              idx.scalar_type() == torch::kLong);
+  TORCH_CHECK(src.is_contiguous() && dst.is_contiguous());
+  const int64_t F = dst.size(1);
+  const int64_t n = idx.numel();
+  TORCH_CHECK(src.size(0) == n && src.size(1) == F);
+  auto stream = current_stream();
+  const int vec = pick_vec(F);
+  if (vec == 4)
+    launch_rowcopy<4, true>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                            dst.data_ptr<float>(), n, F, stream);
+  else if (vec == 2)
+    launch_rowcopy<2, true>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                            dst.data_ptr<float>(), n, F, stream);
+  else
+    launch_rowcopy<1, true>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
+                            dst.data_ptr<float>(), n, F, stream);
+}
+
+void ema_update_hip(torch::Tensor avg, torch::Tensor x, double momentum) {
+  TORCH_CHECK(avg.is_cuda() && x.is_cuda());
+  TORCH_CHECK(avg.is_contiguous() && x.is_contiguous());
+  TORCH_CHECK(avg.numel() == x.numel());
+  const int64_t n = avg.numel();
+  const int threads = 256;
+  int64_t blocks = (n + 4 * threads - 1) / (4 * threads);
+  if (blocks == 0) blocks = 1;
+  hipLaunchKernelGGL(ema_kernel, dim3(blocks), dim3(threads), 0,
+                     current_stream(), avg.data_ptr<float>(),
+                     x.data_ptr<float>(), static_cast<float>(momentum), n);
+  HIP_CHECK(hipGetLastError());
+}

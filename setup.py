@@ -1,0 +1,37 @@
+"""Build the pipegcn_amd native extension in-tree.
+
+Usage:  python setup.py build_ext --inplace
+
+Compiles the C++ host graph core and the gfx950 HIP kernels into
+pipegcn_amd/_C*.so (single offload arch: gfx950 / MI355X — no dual path).
+"""
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+
+ext = CUDAExtension(
+    name="pipegcn_amd._C",
+    sources=[
+        "pipegcn_amd/csrc/bindings.cpp",
+        "pipegcn_amd/csrc/graph_core.cpp",
+        "pipegcn_amd/csrc/hip/kernels.hip",
+    ],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17"],
+    },
+)
+
+setup(
+    name="pipegcn_amd",
+    version="0.1.0",
+    packages=["pipegcn_amd"],
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)
