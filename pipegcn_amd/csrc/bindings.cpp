@@ -28,6 +28,15 @@ static torch::Tensor gather_rows(torch::Tensor src, torch::Tensor idx) {
   return src.index_select(0, idx);
 }
 
+static void gather_rows_out(torch::Tensor src, torch::Tensor idx,
+                            torch::Tensor out) {
+  if (src.is_cuda()) {
+    gather_rows_hip(src, idx, out);
+  } else {
+    torch::index_select_out(out, src, 0, idx);
+  }
+}
+
 static void scatter_add_rows(torch::Tensor dst, torch::Tensor idx,
                              torch::Tensor src) {
   if (dst.is_cuda()) {
@@ -52,6 +61,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "BFS-grown + refined k-way partitioner (cut/vol objective)");
   m.def("spmm", &spmm, "CSR SpMM with fused scale epilogue (fwd & transpose)");
   m.def("gather_rows", &gather_rows, "out[i,:] = src[idx[i],:]");
+  m.def("gather_rows_out", &gather_rows_out,
+        "gather into a preallocated out buffer");
   m.def("scatter_add_rows", &scatter_add_rows, "dst[idx[i],:] += src[i,:]");
   m.def("ema_update", &ema_update, "avg = m*avg + (1-m)*x");
   m.attr("with_hip") = true;
