@@ -66,6 +66,10 @@ def build_runtime_partition(part: PartData,
         nt = torch.empty_like(t)
         nt[new_id] = t
         ndata[k] = nt
+    # reshuffled global id of each (renumbered) inner node — debugging/tests
+    gid = torch.arange(offsets[rank], offsets[rank] + num_in)
+    ndata["gid"] = torch.empty_like(gid)
+    ndata["gid"][new_id] = gid
 
     u, v = part.edges[0], part.edges[1]
     u = torch.where(u < num_in, new_id[u.clamp(max=num_in - 1)], u)

@@ -174,12 +174,16 @@ class Buffer:
             self._check_err()
             if layer == 0:
                 return grad  # input features carry no remote grads
+            if not grad.is_contiguous():  # e.g. expanded grad from .sum()
+                grad = grad.contiguous()
             if not self._pipeline:
                 with comm_timer.timer(f"backward_{layer}"):
                     self._submit_grad(layer, grad)
                     self._wait_host(self._b_host_evt[layer])
                     self._wait_dev(self._b_done_evt[layer])
                 self._apply_grad(layer, grad)
+                if self._use_cuda:
+                    self._b_consumed_evt[layer].record()
             else:
                 if self._epoch > 0:
                     with comm_timer.timer(f"backward_{layer}"):

@@ -212,7 +212,7 @@ def run(part: PartData, args, device: str = "cpu",
     result_file_name = "results/%s_n%d_p%d%s.txt" % (
         args.dataset, args.n_partitions, int(args.enable_pipeline), suffix)
 
-    train_dur, comm_dur, reduce_dur = [], [], []
+    train_dur, comm_dur, reduce_dur, loss_hist = [], [], [], []
     best_model, best_acc = None, 0.0
     thread = None
     pool = ThreadPool(processes=1)
@@ -250,6 +250,7 @@ def run(part: PartData, args, device: str = "cpu",
                       float(np.mean(reduce_dur or [0])),
                       loss.item() / part_train))
         comm_timer.clear()
+        loss_hist.append(loss.item())
         del loss
 
         if (rank == 0 and args.eval
@@ -273,6 +274,7 @@ def run(part: PartData, args, device: str = "cpu",
     ctx.buffer.synchronize()
     summary = {
         "rank": rank,
+        "losses": loss_hist,
         "mean_epoch_s": float(np.mean(train_dur)) if train_dur else None,
         "mean_comm_s": float(np.mean(comm_dur)) if comm_dur else None,
         "mean_reduce_s": float(np.mean(reduce_dur)) if reduce_dur else None,

@@ -1,0 +1,98 @@
+"""Unit tests for the native graph core (CSR build, partitioner, CPU SpMM)."""
+import torch
+
+from pipegcn_amd import native
+from pipegcn_amd.graph.csr import CSR, FullGraph, HaloGraph
+
+
+def random_coo(n_src, n_dst, e, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    u = torch.randint(0, n_src, (e,), generator=g)
+    v = torch.randint(0, n_dst, (e,), generator=g)
+    return u, v
+
+
+def spmm_torch_ref(u, v, feat, num_rows, scale=None):
+    out = torch.zeros(num_rows, feat.shape[1])
+    out.index_add_(0, v, feat[u])
+    if scale is not None:
+        out *= scale.unsqueeze(1)
+    return out
+
+
+def test_build_csr_roundtrip():
+    u, v = random_coo(50, 30, 400)
+    indptr, indices = native().build_csr(u, v, 30)
+    assert indptr[-1].item() == 400
+    # every edge present exactly once
+    got = sorted((indices[e].item(), r) for r in range(30)
+                 for e in range(indptr[r], indptr[r + 1]))
+    want = sorted(zip(u.tolist(), v.tolist()))
+    assert got == want
+
+
+def test_spmm_cpu_matches_torch():
+    for f in (1, 3, 8, 33, 64):
+        u, v = random_coo(40, 25, 300, seed=f)
+        feat = torch.randn(40, f)
+        scale = torch.rand(25) + 0.5
+        csr = CSR.from_coo(u, v, 25, 40)
+        out = native().spmm(csr.indptr, csr.indices, feat, scale, 25)
+        ref = spmm_torch_ref(u, v, feat, 25, scale)
+        assert torch.allclose(out, ref, atol=1e-5), f"F={f}"
+
+
+def test_spmm_no_scale():
+    u, v = random_coo(40, 25, 300)
+    feat = torch.randn(40, 8)
+    csr = CSR.from_coo(u, v, 25, 40)
+    out = native().spmm(csr.indptr, csr.indices, feat, torch.Tensor(), 25)
+    assert torch.allclose(out, spmm_torch_ref(u, v, feat, 25), atol=1e-5)
+
+
+def test_partitioner_balance_and_cut():
+    # community graph: partitioner should find communities
+    n, k = 400, 4
+    g = torch.Generator().manual_seed(3)
+    dst = torch.randint(0, n, (6000,), generator=g)
+    comm = dst // (n // k)
+    local = torch.rand(6000, generator=g) < 0.9
+    src = torch.where(
+        local,
+        comm * (n // k) + torch.randint(0, n // k, (6000,), generator=g),
+        torch.randint(0, n, (6000,), generator=g))
+    su, sv = torch.cat([src, dst]), torch.cat([dst, src])
+    indptr, indices = native().build_csr(su, sv, n)
+    for obj in (0, 1):
+        part = native().partition_graph(indptr, indices, k, obj, 0.05, 8, 0)
+        sizes = torch.bincount(part.long(), minlength=k)
+        assert sizes.min() >= n / k * 0.9, sizes
+        assert sizes.max() <= n / k * 1.1, sizes
+        cut = (part[src.int().long()] != part[dst.long()]).float().mean()
+        assert cut < 0.5, f"cut fraction {cut} too high (random ~0.75)"
+
+
+def test_partitioner_single_part():
+    u, v = random_coo(20, 20, 50)
+    indptr, indices = native().build_csr(u, v, 20)
+    part = native().partition_graph(indptr, indices, 1, 0, 0.05, 4, 0)
+    assert (part == 0).all()
+
+
+def test_halo_graph_csc_is_transpose():
+    u, v = random_coo(35, 20, 200)
+    g = HaloGraph.from_edges(u, v, 20, 35)
+    feat = torch.randn(35, 5)
+    gout = torch.randn(20, 5)
+    # <A x, g> == <x, A^T g>
+    ax = native().spmm(g.csr.indptr, g.csr.indices, feat, torch.Tensor(), 20)
+    atg = native().spmm(g.csc.indptr, g.csc.indices, gout, torch.Tensor(),
+                        35)
+    assert torch.allclose((ax * gout).sum(), (feat * atg).sum(), atol=1e-3)
+
+
+def test_full_graph_degrees():
+    u, v = random_coo(30, 30, 100)
+    fg = FullGraph.from_coo(u, v, 30)
+    ref = torch.bincount(v, minlength=30).float()
+    assert torch.equal(fg.in_degrees(), ref)

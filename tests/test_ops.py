@@ -1,0 +1,123 @@
+"""Numerics tests for the HIP kernels vs plain PyTorch fp32 references.
+
+CPU variants cover the native C++ paths; @pytest.mark.gpu variants run the
+hand-written gfx950 kernels on a real MI355X.
+"""
+import pytest
+import torch
+
+from pipegcn_amd import ops
+from pipegcn_amd.graph.csr import CSR, HaloGraph
+
+
+def make_graph(n_src=300, n_dst=200, e=4000, f=256, seed=0, device="cpu"):
+    g = torch.Generator().manual_seed(seed)
+    u = torch.randint(0, n_src, (e,), generator=g)
+    v = torch.randint(0, n_dst, (e,), generator=g)
+    feat = torch.randn(n_src, f, generator=g)
+    hg = HaloGraph.from_edges(u, v, n_dst, n_src).to(device)
+    return u, v, hg, feat.to(device)
+
+
+def torch_spmm(u, v, feat, num_rows, scale=None):
+    out = torch.zeros(num_rows, feat.shape[1], device=feat.device)
+    out.index_add_(0, v.to(feat.device), feat[u.to(feat.device)])
+    if scale is not None:
+        out *= scale.unsqueeze(1)
+    return out
+
+
+@pytest.mark.parametrize("f", [4, 41, 100, 256, 602])
+def test_spmm_autograd_cpu(f):
+    _spmm_autograd_check("cpu", f)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("f", [4, 41, 100, 256, 602])
+def test_spmm_autograd_gpu(f):
+    _spmm_autograd_check("cuda", f)
+
+
+def _spmm_autograd_check(device, f):
+    u, v, hg, feat = make_graph(f=f, device=device)
+    deg = torch.bincount(v, minlength=hg.num_in).float().clamp(min=1)
+    inv_deg = (1.0 / deg).to(device)
+    feat = feat.requires_grad_(True)
+    out = ops.spmm_mean(hg, feat, inv_deg)
+    # torch fp32 reference with autograd
+    feat_ref = feat.detach().clone().requires_grad_(True)
+    ref = torch_spmm(u, v, feat_ref, hg.num_in, inv_deg.cpu().to(device))
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+    gout = torch.randn_like(out)
+    out.backward(gout)
+    ref.backward(gout)
+    assert torch.allclose(feat.grad, feat_ref.grad, atol=1e-4, rtol=1e-4)
+
+
+@pytest.mark.gpu
+def test_spmm_empty_rows_gpu():
+    # rows with no in-edges must come out zero, not garbage
+    u = torch.tensor([0, 1])
+    v = torch.tensor([0, 0])
+    hg = HaloGraph.from_edges(u, v, 5, 5).to("cuda")
+    feat = torch.randn(5, 7, device="cuda")
+    out = ops.spmm(hg.csr, feat, None)
+    assert torch.allclose(out[1:], torch.zeros(4, 7, device="cuda"))
+    assert torch.allclose(out[0], feat[0] + feat[1])
+
+
+@pytest.mark.parametrize("device", ["cpu"])
+@pytest.mark.parametrize("f", [3, 64, 130])
+def test_gather_scatter_cpu(device, f):
+    _gather_scatter_check(device, f)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("f", [3, 64, 130])
+def test_gather_scatter_gpu(f):
+    _gather_scatter_check("cuda", f)
+
+
+def _gather_scatter_check(device, f):
+    src = torch.randn(100, f, device=device)
+    idx = torch.randperm(100, device=device)[:30]
+    g = ops.gather_rows(src, idx)
+    assert torch.allclose(g, src[idx])
+    out = torch.empty(30, f, device=device)
+    ops.gather_rows_into(src, idx, out)
+    assert torch.allclose(out, src[idx])
+    dst = torch.randn(100, f, device=device)
+    ref = dst.clone()
+    add = torch.randn(30, f, device=device)
+    ops.scatter_add_rows(dst, idx, add)
+    ref[idx] += add
+    assert torch.allclose(dst, ref, atol=1e-5)
+
+
+def test_ema_cpu():
+    _ema_check("cpu")
+
+
+@pytest.mark.gpu
+def test_ema_gpu():
+    _ema_check("cuda")
+
+
+def _ema_check(device):
+    for numel in (12, 1024, 1027):
+        avg = torch.randn(numel, device=device).view(1, -1).contiguous()
+        x = torch.randn(numel, device=device).view(1, -1).contiguous()
+        ref = 0.95 * avg + 0.05 * x
+        ops.ema_update(avg, x, 0.95)
+        assert torch.allclose(avg, ref, atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_native_kernels_loaded_on_gpu():
+    """The GPU path must run OUR extension, not an eager fallback."""
+    import pipegcn_amd
+
+    assert pipegcn_amd.HAS_NATIVE
+    assert pipegcn_amd._C.with_hip
+    # the .so must live in-tree so it ships with the repo snapshot
+    assert "pipegcn_amd" in pipegcn_amd._C.__file__
