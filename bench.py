@@ -1,0 +1,179 @@
+"""Flagship benchmark: full-graph GraphSAGE training epoch time on a
+synthetic Reddit-shaped graph (BASELINE.json metric: "epoch time (s) +
+boundary-comm overlap %, Reddit 3-layer GraphSAGE at 1/2/4/8 parts").
+
+One step == one full-graph training epoch (forward + sum-loss + backward +
+pipelined boundary exchange + gradient all-reduce + Adam step), strong
+scaling: the Reddit-shaped graph is partitioned over N GPUs.
+
+Launch:  python bench.py --gpus N --steps K --warmup W
+  N == 1: runs standalone.
+  N > 1 : launched by the driver as
+          python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+              --master-addr 127.0.0.1 bench.py --gpus N ...
+          (reads RANK/WORLD_SIZE/MASTER_* from the env, RCCL over xGMI).
+
+Rank 0 prints ONE JSON line with the whole-job metric (max epoch time over
+ranks) plus the comm-overlap statistics measured with HIP events on the side
+comm stream.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--shape", type=str, default="reddit",
+                    help="synthetic shape (reddit/ogbn-products/yelp/small)")
+    ap.add_argument("--n-hidden", type=int, default=256)
+    ap.add_argument("--n-layers", type=int, default=3)
+    ap.add_argument("--no-pipeline", action="store_true",
+                    help="disable cross-epoch pipelining (vanilla mode)")
+    ap.add_argument("--device", type=str, default=None)
+    args = ap.parse_args()
+
+    env_world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    world = max(env_world, 1)
+    if world == 1 and args.gpus > 1:
+        raise SystemExit("bench.py with --gpus N>1 must be launched via "
+                         "torch.distributed.run (one rank per GPU)")
+
+    use_cuda = torch.cuda.is_available()
+    device = args.device or ("cuda" if use_cuda else "cpu")
+    backend = "nccl" if device.startswith("cuda") else "gloo"
+    if device.startswith("cuda"):
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29500")
+    dist.init_process_group(backend, rank=rank, world_size=world)
+
+    from pipegcn_amd.graph.halo import build_runtime_partition
+    from pipegcn_amd.graph.synthetic import SHAPES, synth_partition
+    from pipegcn_amd.models.sage import GraphSAGE
+    from pipegcn_amd.parallel import context as ctx
+    from pipegcn_amd.trainer import get_layer_size
+    from pipegcn_amd.utils.timer import comm_timer
+    import torch.nn.functional as F
+
+    torch.manual_seed(0)
+    part = synth_partition(args.shape, rank, world, seed=0)
+    rp = build_runtime_partition(part, device=device)
+
+    pipeline = not args.no_pipeline
+    layer_size = get_layer_size(part.n_feat, args.n_hidden, part.n_class,
+                                args.n_layers)
+    comm_group = dist.new_group(backend=backend) if world > 1 else None
+    ctx.buffer.init_buffer(rp.num_in, rp.num_all, rp.boundary, rp.recv_shape,
+                           layer_size[: args.n_layers], use_pp=False,
+                           backend=backend, pipeline=pipeline,
+                           device=device, group=comm_group,
+                           collect_stats=True)
+
+    model = GraphSAGE(layer_size, F.relu, use_pp=False, dropout=0.5,
+                      norm="layer", n_linear=0,
+                      train_size=part.n_train).to(device)
+    ctx.reducer.init(model)
+    loss_fcn = torch.nn.CrossEntropyLoss(reduction="sum")
+    optimizer = torch.optim.Adam(model.parameters(), lr=0.01)
+
+    feat = rp.ndata["feat"]
+    in_deg = rp.ndata["in_degree"]
+    labels = rp.ndata["label"][: rp.num_train]
+    model.train()
+
+    def step():
+        logits = model(rp.graph, feat, in_deg)
+        loss = loss_fcn(logits[: rp.num_train], labels)
+        optimizer.zero_grad(set_to_none=True)
+        loss.backward()
+        ctx.buffer.next_epoch()
+        ctx.reducer.synchronize(part.n_train)
+        optimizer.step()
+        comm_timer.clear()
+        return loss
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        if device.startswith("cuda"):
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    ctx.buffer.pop_comm_stats()  # reset overlap stats
+    wait_s = 0.0
+    barrier_sync()
+    t0 = time.time()
+    for _ in range(args.steps):
+        logits = model(rp.graph, feat, in_deg)
+        loss = loss_fcn(logits[: rp.num_train], labels)
+        optimizer.zero_grad(set_to_none=True)
+        loss.backward()
+        ctx.buffer.next_epoch()
+        ctx.reducer.synchronize(part.n_train)
+        optimizer.step()
+        wait_s += comm_timer.tot_time()
+        comm_timer.clear()
+    barrier_sync()
+    elapsed = time.time() - t0
+    comm_busy = ctx.buffer.pop_comm_stats()
+
+    # whole-job epoch time = max over ranks
+    t = torch.tensor([elapsed, comm_busy, wait_s], dtype=torch.float64)
+    if world > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed, comm_busy, wait_s = t.tolist()
+    epoch_s = elapsed / args.steps
+    overlap_pct = (100.0 * max(comm_busy - wait_s, 0.0) / comm_busy
+                   if comm_busy > 1e-9 else None)
+
+    if rank == 0:
+        n, avg_deg, n_feat, n_class, _ = SHAPES[args.shape]
+        baseline = 0.2660  # BASELINE.md Reddit epoch time (rank 0), other hw
+        result = {
+            "metric": "epoch time (s), Reddit 3-layer GraphSAGE "
+                      "(full-graph, pipelined)",
+            "value": epoch_s,
+            "unit": "s/epoch",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": epoch_s * 1e3,
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": (epoch_s / baseline
+                            if args.shape == "reddit" else None),
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"graphsage-{args.n_layers}L-h{args.n_hidden}",
+                "graph": args.shape,
+                "num_nodes": n,
+                "avg_degree": avg_deg,
+                "n_feat": n_feat,
+                "n_class": n_class,
+                "pipeline": pipeline,
+                "parallelism": f"graph-partition dp{world}",
+                "comm_busy_s_per_epoch": comm_busy / args.steps,
+                "comm_wait_s_per_epoch": wait_s / args.steps,
+                "boundary_comm_overlap_pct": overlap_pct,
+            },
+        }
+        print(json.dumps(result))
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
