@@ -9,14 +9,14 @@
 #include "common.h"
 
 static torch::Tensor spmm(torch::Tensor indptr, torch::Tensor indices,
-                          torch::Tensor feat, torch::Tensor scale,
-                          int64_t num_rows) {
+                          torch::Tensor feat, torch::Tensor dst_scale,
+                          torch::Tensor src_scale, int64_t num_rows) {
   if (feat.is_cuda()) {
     auto out = torch::empty({num_rows, feat.size(1)}, feat.options());
-    spmm_csr_hip(indptr, indices, feat, scale, out);
+    spmm_csr_hip(indptr, indices, feat, dst_scale, src_scale, out);
     return out;
   }
-  return spmm_cpu(indptr, indices, feat, scale, num_rows);
+  return spmm_cpu(indptr, indices, feat, dst_scale, src_scale, num_rows);
 }
 
 static torch::Tensor gather_rows(torch::Tensor src, torch::Tensor idx) {

@@ -38,8 +38,8 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
 // CPU CSR SpMM: out[r, :] = scale[r] * sum_{c in row r} feat[indices[c], :]
 // scale may be an undefined tensor (no scaling). feat float32 [num_src, F].
 torch::Tensor spmm_cpu(torch::Tensor indptr, torch::Tensor indices,
-                       torch::Tensor feat, torch::Tensor scale,
-                       int64_t num_rows);
+                       torch::Tensor feat, torch::Tensor dst_scale,
+                       torch::Tensor src_scale, int64_t num_rows);
 
 // ---------------------------------------------------------------------------
 // HIP kernel launchers (hip/kernels.hip) — gfx950 only.
@@ -48,7 +48,8 @@ torch::Tensor spmm_cpu(torch::Tensor indptr, torch::Tensor indices,
 // out[r,:] = scale[r] * sum_{e in [indptr[r], indptr[r+1])} feat[indices[e],:]
 // All tensors on device. feat fp32 [num_src, F], out fp32 [num_rows, F].
 void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
-                  torch::Tensor feat, torch::Tensor scale, torch::Tensor out);
+                  torch::Tensor feat, torch::Tensor dst_scale,
+                  torch::Tensor src_scale, torch::Tensor out);
 
 // out[i,:] = src[idx[i],:]
 void gather_rows_hip(torch::Tensor src, torch::Tensor idx, torch::Tensor out);

@@ -57,8 +57,8 @@ std::vector<torch::Tensor> build_csr(torch::Tensor u, torch::Tensor v,
 }
 
 torch::Tensor spmm_cpu(torch::Tensor indptr, torch::Tensor indices,
-                       torch::Tensor feat, torch::Tensor scale,
-                       int64_t num_rows) {
+                       torch::Tensor feat, torch::Tensor dst_scale,
+                       torch::Tensor src_scale, int64_t num_rows) {
   TORCH_CHECK(feat.dim() == 2 && feat.scalar_type() == torch::kFloat,
               "spmm_cpu: feat must be fp32 [N,F]");
   feat = feat.contiguous();
@@ -68,15 +68,20 @@ torch::Tensor spmm_cpu(torch::Tensor indptr, torch::Tensor indices,
   const int32_t* xp = indices.data_ptr<int32_t>();
   const float* fp = feat.data_ptr<float>();
   float* op = out.data_ptr<float>();
-  const bool has_scale = scale.defined() && scale.numel() > 0;
-  const float* sp = has_scale ? scale.contiguous().data_ptr<float>() : nullptr;
+  const bool has_scale = dst_scale.defined() && dst_scale.numel() > 0;
+  auto dsc = has_scale ? dst_scale.contiguous() : dst_scale;
+  const float* sp = has_scale ? dsc.data_ptr<float>() : nullptr;
+  const bool has_src = src_scale.defined() && src_scale.numel() > 0;
+  auto ssc = has_src ? src_scale.contiguous() : src_scale;
+  const float* ssp = has_src ? ssc.data_ptr<float>() : nullptr;
 
   at::parallel_for(0, num_rows, 64, [&](int64_t begin, int64_t end) {
     for (int64_t r = begin; r < end; ++r) {
       float* orow = op + r * F;
       for (int64_t e = ip[r]; e < ip[r + 1]; ++e) {
         const float* frow = fp + static_cast<int64_t>(xp[e]) * F;
-        for (int64_t k = 0; k < F; ++k) orow[k] += frow[k];
+        const float sv = ssp ? ssp[xp[e]] : 1.f;
+        for (int64_t k = 0; k < F; ++k) orow[k] += sv * frow[k];
       }
       if (has_scale) {
         const float s = sp[r];

@@ -20,6 +20,8 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
 #define HIP_CHECK(expr)                                              \
   do {                                                               \
     hipError_t _e = (expr);                                          \
@@ -39,11 +41,19 @@ inline hipStream_t current_stream() {
 // One wave per (row, chunk) pair, grid-stride. VEC in {4, 2, 1}.
 // ---------------------------------------------------------------------------
 
-template <int VEC>
+// CHUNK_OUTER: iterate column panels as the SLOWEST dimension so all
+// resident waves gather from one feature panel (num_src × 64·VEC cols) at a
+// time — the panel stays Infinity-Cache(L3, 256 MiB)-resident and the random
+// neighbor-row gathers become L3 hits instead of HBM reads. With chunk
+// fastest (CHUNK_OUTER=false) every panel streams concurrently and the
+// combined working set thrashes L3.  HAS_SRC_SCALE fuses a per-source-row
+// scale (the transpose/backward SpMM's D^{-1} pre-scale) into the gather.
+template <int VEC, bool CHUNK_OUTER, bool HAS_SRC_SCALE>
 __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
                                 const int32_t* __restrict__ indices,
                                 const float* __restrict__ feat,
-                                const float* __restrict__ scale,
+                                const float* __restrict__ dst_scale,
+                                const float* __restrict__ src_scale,
                                 float* __restrict__ out, int64_t num_rows,
                                 int64_t F, int64_t nchunks) {
   const int64_t wave_global =
@@ -54,8 +64,14 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
   const int64_t npairs = num_rows * nchunks;
 
   for (int64_t pair = wave_global; pair < npairs; pair += nwaves) {
-    const int64_t r = pair / nchunks;
-    const int64_t chunk = pair % nchunks;
+    int64_t r, chunk;
+    if (CHUNK_OUTER) {
+      r = pair % num_rows;
+      chunk = pair / num_rows;
+    } else {
+      r = pair / nchunks;
+      chunk = pair % nchunks;
+    }
     const int64_t f0 = chunk * (kWave * VEC) + lane * VEC;
     if (f0 >= F) continue;
     const bool full = (f0 + VEC <= F);
@@ -75,6 +91,13 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
         const int64_t u1 = indices[e + 1];
         const int64_t u2 = indices[e + 2];
         const int64_t u3 = indices[e + 3];
+        float s0 = 1.f, s1 = 1.f, s2 = 1.f, s3 = 1.f;
+        if (HAS_SRC_SCALE) {
+          s0 = src_scale[u0];
+          s1 = src_scale[u1];
+          s2 = src_scale[u2];
+          s3 = src_scale[u3];
+        }
         float v0[VEC], v1[VEC], v2[VEC], v3[VEC];
 #pragma unroll
         for (int k = 0; k < VEC; ++k) v0[k] = feat[u0 * F + f0 + k];
@@ -84,25 +107,34 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
         for (int k = 0; k < VEC; ++k) v2[k] = feat[u2 * F + f0 + k];
 #pragma unroll
         for (int k = 0; k < VEC; ++k) v3[k] = feat[u3 * F + f0 + k];
+        if (HAS_SRC_SCALE) {
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) acc[k] += v0[k] + v1[k] + v2[k] + v3[k];
+          for (int k = 0; k < VEC; ++k)
+            acc[k] += s0 * v0[k] + s1 * v1[k] + s2 * v2[k] + s3 * v3[k];
+        } else {
+#pragma unroll
+          for (int k = 0; k < VEC; ++k)
+            acc[k] += v0[k] + v1[k] + v2[k] + v3[k];
+        }
       }
       for (; e < e_end; ++e) {
         const int64_t u = indices[e];
+        const float s = HAS_SRC_SCALE ? src_scale[u] : 1.f;
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) acc[k] += feat[u * F + f0 + k];
+        for (int k = 0; k < VEC; ++k) acc[k] += s * feat[u * F + f0 + k];
       }
-      const float s = scale ? scale[r] : 1.f;
+      const float s = dst_scale ? dst_scale[r] : 1.f;
 #pragma unroll
       for (int k = 0; k < VEC; ++k) out[r * F + f0 + k] = acc[k] * s;
     } else {
       // ragged tail chunk: scalar guarded
       for (; e < e_end; ++e) {
         const int64_t u = indices[e];
+        const float s = HAS_SRC_SCALE ? src_scale[u] : 1.f;
         for (int k = 0; k < VEC && f0 + k < F; ++k)
-          acc[k] += feat[u * F + f0 + k];
+          acc[k] += s * feat[u * F + f0 + k];
       }
-      const float s = scale ? scale[r] : 1.f;
+      const float s = dst_scale ? dst_scale[r] : 1.f;
       for (int k = 0; k < VEC && f0 + k < F; ++k)
         out[r * F + f0 + k] = acc[k] * s;
     }
@@ -111,18 +143,35 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
 
 template <int VEC>
 void launch_spmm(const int64_t* indptr, const int32_t* indices,
-                 const float* feat, const float* scale, float* out,
-                 int64_t num_rows, int64_t F, hipStream_t stream) {
+                 const float* feat, const float* dst_scale,
+                 const float* src_scale, float* out, int64_t num_src,
+                 int64_t num_rows, int64_t F, bool chunk_outer,
+                 hipStream_t stream) {
   const int64_t nchunks = (F + kWave * VEC - 1) / (kWave * VEC);
   const int threads = 256;  // 4 waves
   const int64_t npairs = num_rows * nchunks;
   int64_t blocks = (npairs * kWave + threads - 1) / threads;
-  // MI355X: 256 CUs; >= 2048 workgroups fills the chip with headroom.
-  blocks = std::min<int64_t>(blocks, 8 * 65536);
+  // MI355X: 256 CUs; cap the grid, grid-stride covers the rest. For
+  // chunk-outer panel locality the wave front must sweep rows in order, so
+  // cap at full residency (256 CU × 8 blocks ≈ upper bound).
+  blocks = std::min<int64_t>(blocks, chunk_outer ? 4096 : 8 * 65536);
   if (blocks == 0) blocks = 1;
-  hipLaunchKernelGGL(HIP_KERNEL_NAME(spmm_csr_kernel<VEC>), dim3(blocks),
-                     dim3(threads), 0, stream, indptr, indices, feat, scale,
-                     out, num_rows, F, nchunks);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, stream, indptr,
+                       indices, feat, dst_scale, src_scale, out, num_rows, F,
+                       nchunks);
+  };
+  if (chunk_outer) {
+    if (src_scale)
+      launch(HIP_KERNEL_NAME(spmm_csr_kernel<VEC, true, true>));
+    else
+      launch(HIP_KERNEL_NAME(spmm_csr_kernel<VEC, true, false>));
+  } else {
+    if (src_scale)
+      launch(HIP_KERNEL_NAME(spmm_csr_kernel<VEC, false, true>));
+    else
+      launch(HIP_KERNEL_NAME(spmm_csr_kernel<VEC, false, false>));
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -200,43 +249,66 @@ __global__ void ema_kernel(float* __restrict__ avg,
   }
 }
 
-int pick_vec(int64_t F) {
-  if (F % 4 == 0) return 4;
-  if (F % 2 == 0) return 2;
-  return 1;
+int pick_vec(int64_t F, int64_t num_src) {
+  // Prefer the widest load whose column panel (num_src × 64·VEC × 4 B) stays
+  // L3-resident (256 MiB Infinity Cache; target ≤ ~160 MiB leaves room for
+  // the streaming index array). Overridable for A/B benchmarking.
+  if (const char* e = std::getenv("PIPEGCN_SPMM_VEC")) {
+    int v = std::atoi(e);
+    if ((v == 4 || v == 2 || v == 1) && F % v == 0) return v;
+  }
+  const int64_t target = 160ll << 20;
+  int best_fit = 0, best_any = 0;
+  for (int v : {4, 2, 1}) {
+    if (F % v != 0) continue;
+    if (!best_any) best_any = v;
+    const int64_t slab = num_src * std::min<int64_t>(64 * v, F) * 4;
+    if (!best_fit && slab <= target) best_fit = v;
+  }
+  return best_fit ? best_fit : best_any;
 }
 
 }  // namespace
 
 void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
-                  torch::Tensor feat, torch::Tensor scale, torch::Tensor out) {
+                  torch::Tensor feat, torch::Tensor dst_scale,
+                  torch::Tensor src_scale, torch::Tensor out) {
   TORCH_CHECK(feat.is_cuda() && out.is_cuda(), "spmm_csr_hip: device tensors");
   TORCH_CHECK(feat.scalar_type() == torch::kFloat, "fp32 only");
   TORCH_CHECK(feat.is_contiguous() && out.is_contiguous());
   TORCH_CHECK(indptr.scalar_type() == torch::kLong &&
               indices.scalar_type() == torch::kInt);
   const int64_t num_rows = out.size(0);
+  const int64_t num_src = feat.size(0);
   const int64_t F = feat.size(1);
   TORCH_CHECK(out.size(1) == F && indptr.numel() == num_rows + 1);
-  const float* sp = nullptr;
-  if (scale.defined() && scale.numel() > 0) {
-    TORCH_CHECK(scale.is_contiguous() && scale.numel() == num_rows);
-    sp = scale.data_ptr<float>();
+  const float* dsp = nullptr;
+  if (dst_scale.defined() && dst_scale.numel() > 0) {
+    TORCH_CHECK(dst_scale.is_contiguous() && dst_scale.numel() == num_rows);
+    dsp = dst_scale.data_ptr<float>();
+  }
+  const float* ssp = nullptr;
+  if (src_scale.defined() && src_scale.numel() > 0) {
+    TORCH_CHECK(src_scale.is_contiguous() && src_scale.numel() == num_src);
+    ssp = src_scale.data_ptr<float>();
   }
   auto stream = current_stream();
-  const int vec = pick_vec(F);
+  const int vec = pick_vec(F, num_src);
+  bool chunk_outer = (F + 64 * vec - 1) / (64 * vec) > 1;
+  if (const char* e = std::getenv("PIPEGCN_SPMM_ORDER"))
+    chunk_outer = (e[0] == 'o');
   if (vec == 4)
     launch_spmm<4>(indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
-                   feat.data_ptr<float>(), sp, out.data_ptr<float>(), num_rows,
-                   F, stream);
+                   feat.data_ptr<float>(), dsp, ssp, out.data_ptr<float>(),
+                   num_src, num_rows, F, chunk_outer, stream);
   else if (vec == 2)
     launch_spmm<2>(indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
-                   feat.data_ptr<float>(), sp, out.data_ptr<float>(), num_rows,
-                   F, stream);
+                   feat.data_ptr<float>(), dsp, ssp, out.data_ptr<float>(),
+                   num_src, num_rows, F, chunk_outer, stream);
   else
     launch_spmm<1>(indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
-                   feat.data_ptr<float>(), sp, out.data_ptr<float>(), num_rows,
-                   F, stream);
+                   feat.data_ptr<float>(), dsp, ssp, out.data_ptr<float>(),
+                   num_src, num_rows, F, chunk_outer, stream);
 }
 
 void gather_rows_hip(torch::Tensor src, torch::Tensor idx, torch::Tensor out) {
@@ -248,7 +320,7 @@ void gather_rows_hip(torch::Tensor src, torch::Tensor idx, torch::Tensor out) {
   const int64_t n = idx.numel();
   TORCH_CHECK(out.size(0) == n && out.size(1) == F);
   auto stream = current_stream();
-  const int vec = pick_vec(F);
+  const int vec = pick_vec(F, 0);
   if (vec == 4)
     launch_rowcopy<4, false>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
                              out.data_ptr<float>(), n, F, stream);
@@ -272,7 +344,7 @@ void scatter_add_rows_hip(torch::Tensor dst, torch::Tensor idx,
   const int64_t n = idx.numel();
   TORCH_CHECK(src.size(0) == n && src.size(1) == F);
   auto stream = current_stream();
-  const int vec = pick_vec(F);
+  const int vec = pick_vec(F, 0);
   if (vec == 4)
     launch_rowcopy<4, true>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
                             dst.data_ptr<float>(), n, F, stream);

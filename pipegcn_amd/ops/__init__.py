@@ -19,10 +19,12 @@ from pipegcn_amd.graph.csr import CSR, HaloGraph
 
 
 def spmm(csr: CSR, feat: torch.Tensor,
-         scale: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """out[r,:] = scale[r] * sum_{u in N(r)} feat[u,:]  (no autograd)."""
+         scale: Optional[torch.Tensor] = None,
+         src_scale: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out[r,:] = scale[r] * sum_{u in N(r)} src_scale[u] * feat[u,:]."""
     s = scale if scale is not None else torch.Tensor()
-    return native().spmm(csr.indptr, csr.indices, feat.contiguous(), s,
+    ss = src_scale if src_scale is not None else torch.Tensor()
+    return native().spmm(csr.indptr, csr.indices, feat.contiguous(), s, ss,
                          csr.num_rows)
 
 
@@ -40,8 +42,9 @@ class _SpmmMean(torch.autograd.Function):
     def backward(ctx, grad_out: torch.Tensor):
         (inv_deg,) = ctx.saved_tensors
         g = ctx.graph
-        scaled = grad_out * inv_deg.unsqueeze(1)
-        grad_feat = spmm(g.csc, scaled, None)
+        # transpose SpMM with the D^{-1} pre-scale fused as a source scale
+        grad_feat = spmm(g.csc, grad_out.contiguous(), None,
+                         src_scale=inv_deg)
         return None, grad_feat, None
 
 

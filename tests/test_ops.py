@@ -121,3 +121,20 @@ def test_native_kernels_loaded_on_gpu():
     assert pipegcn_amd._C.with_hip
     # the .so must live in-tree so it ships with the repo snapshot
     assert "pipegcn_amd" in pipegcn_amd._C.__file__
+
+
+def test_spmm_src_scale_cpu():
+    _src_scale_check("cpu")
+
+
+@pytest.mark.gpu
+def test_spmm_src_scale_gpu():
+    _src_scale_check("cuda")
+
+
+def _src_scale_check(device):
+    u, v, hg, feat = make_graph(f=64, device=device)
+    ss = (torch.rand(300) + 0.5).to(device)
+    out = ops.spmm(hg.csr, feat, None, src_scale=ss)
+    ref = torch_spmm(u, v, feat * ss.unsqueeze(1), hg.num_in)
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
