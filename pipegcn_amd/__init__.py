@@ -20,6 +20,8 @@ CLI and checkpoint format follow the reference exactly
 
 __version__ = "0.1.0"
 
+import torch  # noqa: F401  (must load libtorch before the extension)
+
 try:
     from pipegcn_amd import _C  # noqa: F401
 
