@@ -85,12 +85,14 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
     int64_t e = e_begin;
 
     if (full) {
-      // main path: unroll 4 edges so 4 gathers are in flight per lane
+      // main path: unroll 4 edges so 4 gathers are in flight per lane.
+      // Index loads and output stores are NON-TEMPORAL: they are streaming
+      // traffic and must not evict the L3-resident feature panel.
       for (; e + 4 <= e_end; e += 4) {
-        const int64_t u0 = indices[e + 0];
-        const int64_t u1 = indices[e + 1];
-        const int64_t u2 = indices[e + 2];
-        const int64_t u3 = indices[e + 3];
+        const int64_t u0 = __builtin_nontemporal_load(indices + e + 0);
+        const int64_t u1 = __builtin_nontemporal_load(indices + e + 1);
+        const int64_t u2 = __builtin_nontemporal_load(indices + e + 2);
+        const int64_t u3 = __builtin_nontemporal_load(indices + e + 3);
         float s0 = 1.f, s1 = 1.f, s2 = 1.f, s3 = 1.f;
         if (HAS_SRC_SCALE) {
           s0 = src_scale[u0];
@@ -125,7 +127,8 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
       }
       const float s = dst_scale ? dst_scale[r] : 1.f;
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) out[r * F + f0 + k] = acc[k] * s;
+      for (int k = 0; k < VEC; ++k)
+        __builtin_nontemporal_store(acc[k] * s, out + r * F + f0 + k);
     } else {
       // ragged tail chunk: scalar guarded
       for (; e < e_end; ++e) {
@@ -253,19 +256,23 @@ int pick_vec(int64_t F, int64_t num_src) {
   // Prefer the widest load whose column panel (num_src × 64·VEC × 4 B) stays
   // L3-resident (256 MiB Infinity Cache; target ≤ ~160 MiB leaves room for
   // the streaming index array). Overridable for A/B benchmarking.
+  // Measured on MI355X (profiles/spmm_ab.md): the gather path tops out at
+  // ~6.5-7 TB/s logical regardless of L3 panel residency; what matters is
+  // pair-level load balance — prefer the VEC giving ~3-8 column chunks,
+  // else the widest aligned load. (num_src kept for future blocking.)
+  (void)num_src;
   if (const char* e = std::getenv("PIPEGCN_SPMM_VEC")) {
     int v = std::atoi(e);
     if ((v == 4 || v == 2 || v == 1) && F % v == 0) return v;
   }
-  const int64_t target = 160ll << 20;
-  int best_fit = 0, best_any = 0;
+  int best_any = 0;
   for (int v : {4, 2, 1}) {
     if (F % v != 0) continue;
     if (!best_any) best_any = v;
-    const int64_t slab = num_src * std::min<int64_t>(64 * v, F) * 4;
-    if (!best_fit && slab <= target) best_fit = v;
+    const int64_t nchunks = (F + 64 * v - 1) / (64 * v);
+    if (nchunks >= 3 && nchunks <= 8) return v;
   }
-  return best_fit ? best_fit : best_any;
+  return best_any;
 }
 
 }  // namespace
@@ -294,7 +301,7 @@ void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
   }
   auto stream = current_stream();
   const int vec = pick_vec(F, num_src);
-  bool chunk_outer = (F + 64 * vec - 1) / (64 * vec) > 1;
+  bool chunk_outer = false;  // measured: chunk-inner wins at every shape
   if (const char* e = std::getenv("PIPEGCN_SPMM_ORDER"))
     chunk_outer = (e[0] == 'o');
   if (vec == 4)

@@ -55,12 +55,16 @@ class RingTransport:
         """
         if self.size == 1:
             return
-        if self.is_nccl:
-            self._nccl_all_to_all(send, recv)
+        on_cuda = any(t is not None and t.is_cuda for t in recv)
+        if self.is_nccl or not on_cuda:
+            # one batched group (RCCL: a single ncclGroupStart/End over all
+            # ring peers). The SAME branch runs for gloo/CPU so the CPU test
+            # tier exercises the op ordering used on the GPU.
+            self._batched_all_to_all(send, recv)
         else:
             self._gloo_all_to_all(send, recv, key, tag)
 
-    def _nccl_all_to_all(self, send, recv) -> None:
+    def _batched_all_to_all(self, send, recv) -> None:
         ops = []
         for left, right in ring_peers(self.rank, self.size):
             r = recv[left]
