@@ -46,6 +46,20 @@ static void scatter_add_rows(torch::Tensor dst, torch::Tensor idx,
   }
 }
 
+static torch::Tensor sage_dual_gemm(torch::Tensor x1, torch::Tensor x2,
+                                    torch::Tensor w1, torch::Tensor w2,
+                                    torch::Tensor bias) {
+  if (x1.is_cuda()) {
+    auto out = torch::empty({x1.size(0), w1.size(0)}, x1.options());
+    sage_dual_gemm_hip(x1, x2, w1, w2, bias, out);
+    return out;
+  }
+  auto out = torch::mm(x1, w1.t());
+  out.addmm_(x2, w2.t());
+  if (bias.defined() && bias.numel() > 0) out.add_(bias);
+  return out;
+}
+
 static void ema_update(torch::Tensor avg, torch::Tensor x, double momentum) {
   if (avg.is_cuda()) {
     ema_update_hip(avg, x, momentum);
@@ -65,5 +79,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "gather into a preallocated out buffer");
   m.def("scatter_add_rows", &scatter_add_rows, "dst[idx[i],:] += src[i,:]");
   m.def("ema_update", &ema_update, "avg = m*avg + (1-m)*x");
+  m.def("sage_dual_gemm", &sage_dual_gemm,
+        "out = x1 @ w1^T + x2 @ w2^T + bias (MFMA fp32, fused)");
   m.attr("with_hip") = true;
 }

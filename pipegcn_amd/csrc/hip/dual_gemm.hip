@@ -1,0 +1,186 @@
+// Fused dual GEMM for the GraphSAGE layer forward (gfx950 MFMA, exact fp32):
+//
+//     out[M,N] = x1[M,K] @ w1^T + x2[M,K] @ w2^T + bias[N]
+//
+// replaces the reference's two cuBLAS GEMMs + elementwise add
+// (`linear1(feat[:num_dst]) + linear2(ah)`, /root/reference/module/layer.py:51)
+// with ONE kernel: the two products share the accumulator (mathematically a
+// single GEMM over a 2K-long concatenated K axis, without materializing the
+// concat), and the bias add is fused into the epilogue — one output pass
+// instead of three.
+//
+// Uses v_mfma_f32_32x32x2_f32 (f32 in / f32 accumulate — bitwise an fmaf
+// chain, NO TF32 anywhere; see cdna_hip_programming.md §3): 155 TF chip peak.
+// Structure: 128x128 block tile, 4 waves (2x2), each wave a 64x64 tile of
+// 2x2 32x32 fragments; K staged through LDS in 32-deep tiles, +1-dword row
+// padding for conflict-free ds_reads. Correctness-first single-buffer
+// staging — the kernel is MFMA-issue-bound at fp32 rate, so deep staging
+// pipelines buy little here (unlike bf16).
+//
+// Weights are in torch Linear layout [N, K] row-major; they are tiny
+// (<= 616 KB) and L2-resident, so the transposed LDS fill reads them
+// scattered without penalty.
+
+#include "../common.h"
+
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+#include <hip/hip_runtime.h>
+
+namespace {
+
+using f32x16 = __attribute__((__vector_size__(16 * sizeof(float)))) float;
+
+constexpr int BM = 128;
+constexpr int BN = 128;
+constexpr int BK = 32;
+constexpr int PAD = 1;
+
+__global__ __launch_bounds__(256) void sage_dual_gemm_kernel(
+    const float* __restrict__ x1, const float* __restrict__ x2,
+    const float* __restrict__ w1, const float* __restrict__ w2,
+    const float* __restrict__ bias, float* __restrict__ out, int64_t M,
+    int64_t N, int64_t K) {
+  __shared__ float a_lds[BM][BK + PAD];
+  __shared__ float b_lds[BK][BN + 4];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;       // 0..3
+  const int wr = wave >> 1;        // wave row 0..1  (64 rows each)
+  const int wc = wave & 1;         // wave col 0..1  (64 cols each)
+
+  const int64_t m0 = static_cast<int64_t>(blockIdx.x) * BM;
+  const int64_t n0 = static_cast<int64_t>(blockIdx.y) * BN;
+
+  f32x16 acc[2][2] = {};
+
+  const int l31 = lane & 31;
+  const int lk = lane >> 5;  // 0/1 : k within the MFMA K=2 step
+
+  for (int seg = 0; seg < 2; ++seg) {
+    const float* X = seg == 0 ? x1 : x2;
+    const float* W = seg == 0 ? w1 : w2;
+    for (int64_t k0 = 0; k0 < K; k0 += BK) {
+      // ---- stage A tile [BM][BK]: 256 threads x 16 elem = 4 x float4.
+      // thread t covers row = 8 rows/iter pattern: t/8 row, (t%8)*4 col
+      {
+        const int ar = tid >> 3;          // 0..31
+        const int ac = (tid & 7) * 4;     // 0..28
+#pragma unroll
+        for (int rr = 0; rr < 4; ++rr) {
+          const int row = ar + rr * 32;
+          const int64_t gm = m0 + row;
+          float4 val = {0.f, 0.f, 0.f, 0.f};
+          if (gm < M) {
+            const int64_t gk = k0 + ac;
+            if (gk + 3 < K) {
+              val = *reinterpret_cast<const float4*>(X + gm * K + gk);
+            } else {
+              float tmp[4] = {0.f, 0.f, 0.f, 0.f};
+              for (int q = 0; q < 4 && gk + q < K; ++q)
+                tmp[q] = X[gm * K + gk + q];
+              val = {tmp[0], tmp[1], tmp[2], tmp[3]};
+            }
+          }
+          a_lds[row][ac + 0] = val.x;
+          a_lds[row][ac + 1] = val.y;
+          a_lds[row][ac + 2] = val.z;
+          a_lds[row][ac + 3] = val.w;
+        }
+      }
+      // ---- stage B tile [BK][BN] = W[n0:n0+BN][k0:k0+BK] transposed.
+      // 256 threads cover 128 n-rows x 32 k-cols: t/2 = n, (t%2)*16 = k base
+      {
+        const int bn = tid >> 1;           // 0..127
+        const int bk = (tid & 1) * 16;     // 0 or 16
+        const int64_t gn = n0 + bn;
+#pragma unroll
+        for (int q = 0; q < 16; q += 4) {
+          const int64_t gk = k0 + bk + q;
+          float4 val = {0.f, 0.f, 0.f, 0.f};
+          if (gn < N) {
+            if (gk + 3 < K) {
+              val = *reinterpret_cast<const float4*>(W + gn * K + gk);
+            } else {
+              float tmp[4] = {0.f, 0.f, 0.f, 0.f};
+              for (int p = 0; p < 4 && gk + p < K; ++p)
+                tmp[p] = W[gn * K + gk + p];
+              val = {tmp[0], tmp[1], tmp[2], tmp[3]};
+            }
+          }
+          b_lds[bk + q + 0][bn] = val.x;
+          b_lds[bk + q + 1][bn] = val.y;
+          b_lds[bk + q + 2][bn] = val.z;
+          b_lds[bk + q + 3][bn] = val.w;
+        }
+      }
+      __syncthreads();
+
+      // ---- MFMA inner loop: 16 K-steps of 2
+#pragma unroll
+      for (int kk = 0; kk < BK; kk += 2) {
+        float a0 = a_lds[wr * 64 + l31][kk + lk];
+        float a1 = a_lds[wr * 64 + 32 + l31][kk + lk];
+        float b0 = b_lds[kk + lk][wc * 64 + l31];
+        float b1 = b_lds[kk + lk][wc * 64 + 32 + l31];
+        acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0],
+                                                         0, 0, 0);
+        acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1],
+                                                         0, 0, 0);
+        acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0],
+                                                         0, 0, 0);
+        acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1],
+                                                         0, 0, 0);
+      }
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: C/D layout for 32x32 shapes (shape-determined):
+  // col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+#pragma unroll
+  for (int fi = 0; fi < 2; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 2; ++fj) {
+      const int64_t col = n0 + wc * 64 + fj * 32 + l31;
+      if (col >= N) continue;
+      const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int row_in_frag = (reg & 3) + 8 * (reg >> 2) + 4 * lk;
+        const int64_t row = m0 + wr * 64 + fi * 32 + row_in_frag;
+        if (row < M) out[row * N + col] = acc[fi][fj][reg] + bv;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void sage_dual_gemm_hip(torch::Tensor x1, torch::Tensor x2, torch::Tensor w1,
+                        torch::Tensor w2, torch::Tensor bias,
+                        torch::Tensor out) {
+  TORCH_CHECK(x1.is_cuda() && x2.is_cuda() && w1.is_cuda() && w2.is_cuda());
+  TORCH_CHECK(x1.scalar_type() == torch::kFloat, "fp32 only");
+  TORCH_CHECK(x1.is_contiguous() && x2.is_contiguous() &&
+              w1.is_contiguous() && w2.is_contiguous() && out.is_contiguous());
+  const int64_t M = x1.size(0);
+  const int64_t K = x1.size(1);
+  const int64_t N = w1.size(0);
+  TORCH_CHECK(x2.size(0) == M && x2.size(1) == K);
+  TORCH_CHECK(w1.size(1) == K && w2.size(0) == N && w2.size(1) == K);
+  TORCH_CHECK(out.size(0) == M && out.size(1) == N);
+  const float* bp = nullptr;
+  if (bias.defined() && bias.numel() > 0) {
+    TORCH_CHECK(bias.is_contiguous() && bias.numel() == N);
+    bp = bias.data_ptr<float>();
+  }
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  hipLaunchKernelGGL(sage_dual_gemm_kernel, grid, dim3(256), 0, stream,
+                     x1.data_ptr<float>(), x2.data_ptr<float>(),
+                     w1.data_ptr<float>(), w2.data_ptr<float>(), bp,
+                     out.data_ptr<float>(), M, N, K);
+  hipError_t e = hipGetLastError();
+  TORCH_CHECK(e == hipSuccess, "HIP error: ", hipGetErrorString(e));
+}

@@ -55,7 +55,12 @@ class GraphSAGELayer(nn.Module):
             assert isinstance(graph, HaloGraph)
             inv_deg = (1.0 / in_deg.clamp(min=1.0)).contiguous()
             ah = ops.spmm_mean(graph, feat, inv_deg)
-            return self.linear1(feat[: graph.num_in]) + self.linear2(ah)
+            x1 = feat[: graph.num_in]
+            if feat.is_cuda:
+                # hand-written MFMA fused dual GEMM (one output pass)
+                return ops.sage_dual_linear(x1, ah, self.linear1,
+                                            self.linear2)
+            return self.linear1(x1) + self.linear2(ah)
         # eval: full homogeneous graph, degrees from the graph itself
         assert isinstance(graph, FullGraph) and in_deg is None
         degs = graph.in_degrees().clamp(min=1.0)

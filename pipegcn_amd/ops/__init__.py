@@ -80,3 +80,37 @@ def scatter_add_rows(dst: torch.Tensor, idx: torch.Tensor,
 def ema_update(avg: torch.Tensor, x: torch.Tensor, momentum: float) -> None:
     """avg = momentum * avg + (1 - momentum) * x (in place)."""
     native().ema_update(avg, x, momentum)
+
+
+class _SageDualLinear(torch.autograd.Function):
+    """out = x1 @ w1^T + x2 @ w2^T + (b1 + b2).
+
+    Forward runs the hand-written gfx950 MFMA fused dual-GEMM
+    (csrc/hip/dual_gemm.hip); backward uses rocBLAS GEMMs (dgrad/wgrad are
+    plain library shapes).
+    """
+
+    @staticmethod
+    def forward(ctx, x1, x2, w1, w2, b1, b2):
+        ctx.save_for_backward(x1, x2, w1, w2)
+        bias = (b1 + b2) if b1 is not None else torch.Tensor()
+        return native().sage_dual_gemm(x1.contiguous(), x2.contiguous(),
+                                       w1.contiguous(), w2.contiguous(),
+                                       bias)
+
+    @staticmethod
+    def backward(ctx, g):
+        x1, x2, w1, w2 = ctx.saved_tensors
+        g = g.contiguous()
+        gx1 = g @ w1
+        gx2 = g @ w2
+        gw1 = g.t() @ x1
+        gw2 = g.t() @ x2
+        gb = g.sum(0)
+        return gx1, gx2, gw1, gw2, gb, gb
+
+
+def sage_dual_linear(x1, x2, lin1, lin2):
+    """Fused linear1(x1) + linear2(x2) for the SAGE layer."""
+    return _SageDualLinear.apply(x1, x2, lin1.weight, lin2.weight, lin1.bias,
+                                 lin2.bias)

@@ -21,6 +21,7 @@ ext = CUDAExtension(
         "pipegcn_amd/csrc/bindings.cpp",
         "pipegcn_amd/csrc/graph_core.cpp",
         "pipegcn_amd/csrc/hip/kernels.hip",
+        "pipegcn_amd/csrc/hip/dual_gemm.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -138,3 +138,30 @@ def _src_scale_check(device):
     out = ops.spmm(hg.csr, feat, None, src_scale=ss)
     ref = torch_spmm(u, v, feat * ss.unsqueeze(1), hg.num_in)
     assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("m,k,n", [(1000, 602, 256), (777, 256, 256),
+                                   (555, 256, 41), (130, 33, 17)])
+def test_sage_dual_gemm_gpu(m, k, n):
+    torch.manual_seed(m + k + n)
+    x1 = torch.randn(m, k, device="cuda", requires_grad=True)
+    x2 = torch.randn(m, k, device="cuda", requires_grad=True)
+    l1 = torch.nn.Linear(k, n).cuda()
+    l2 = torch.nn.Linear(k, n).cuda()
+    out = ops.sage_dual_linear(x1, x2, l1, l2)
+    ref = l1(x1) + l2(x2)
+    scale = ref.abs().max()
+    assert (out - ref).abs().max() / scale < 1e-5, \
+        (out - ref).abs().max().item()
+    g = torch.randn_like(out)
+    out.backward(g, retain_graph=False)
+    gx1, gx2 = x1.grad.clone(), x2.grad.clone()
+    gw1 = l1.weight.grad.clone()
+    x1.grad = x2.grad = None
+    l1.weight.grad = l1.bias.grad = None
+    ref = l1(x1) + l2(x2)
+    ref.backward(g)
+    assert torch.allclose(gx1, x1.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(gx2, x2.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(gw1, l1.weight.grad, atol=1e-2, rtol=1e-3)
