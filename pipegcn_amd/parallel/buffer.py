@@ -132,6 +132,14 @@ class Buffer:
             self._f_consumed_evt = [None] * L
             self._b_consumed_evt = [None] * L
 
+        # Force collective creation of the RCCL communicator NOW, on the
+        # main thread, at a deterministic point on every rank — before the
+        # comm thread issues its first grouped send/recv concurrently with
+        # main-thread collectives on the default group.
+        if (self._transport is not None and self._use_cuda
+                and self._transport.is_nccl):
+            dist.all_reduce(torch.zeros(1, device=self._device), group=group)
+
         # serialized comm thread
         self._queue: queue.Queue = queue.Queue()
         self._exc = None
