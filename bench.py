@@ -38,6 +38,8 @@ def main():
     ap.add_argument("--no-pipeline", action="store_true",
                     help="disable cross-epoch pipelining (vanilla mode)")
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--backend", type=str, default=None,
+                    help="override torch.distributed backend (tests)")
     args = ap.parse_args()
 
     env_world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -49,7 +51,8 @@ def main():
 
     use_cuda = torch.cuda.is_available()
     device = args.device or ("cuda" if use_cuda else "cpu")
-    backend = "nccl" if device.startswith("cuda") else "gloo"
+    backend = args.backend or ("nccl" if device.startswith("cuda")
+                               else "gloo")
     if device.startswith("cuda"):
         local_rank = int(os.environ.get("LOCAL_RANK", "0"))
         torch.cuda.set_device(local_rank)
