@@ -37,7 +37,10 @@ def run_distributed(fn, world_size, args=(), backend="gloo", timeout=120):
 
     port = free_port()
     ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
+    q = ctx.Queue()
+
+    import queue as _q
+    import time as _t
 
     ps = []
     for rank in range(world_size):
@@ -46,15 +49,32 @@ def run_distributed(fn, world_size, args=(), backend="gloo", timeout=120):
         p.start()
         ps.append(p)
     results = {}
-    for _ in range(world_size):
-        rank, ok, payload = q.get()
-        if not ok:
-            for p in ps:
+    deadline = _t.time() + timeout
+    try:
+        for _ in range(world_size):
+            while True:
+                try:
+                    rank, ok, payload = q.get(timeout=5)
+                    break
+                except _q.Empty:
+                    dead = [p for p in ps if not p.is_alive()
+                            and p.exitcode not in (0, None)]
+                    if dead:
+                        raise RuntimeError(
+                            f"worker died with exitcode "
+                            f"{[p.exitcode for p in dead]}")
+                    if _t.time() > deadline:
+                        raise RuntimeError(
+                            f"distributed test hung (> {timeout}s)")
+            if not ok:
+                raise RuntimeError(f"rank {rank} failed:\n{payload}")
+            results[rank] = payload
+        for p in ps:
+            p.join(30)
+    finally:
+        for p in ps:
+            if p.is_alive():
                 p.terminate()
-            raise RuntimeError(f"rank {rank} failed:\n{payload}")
-        results[rank] = payload
-    for p in ps:
-        p.join(timeout)
     return [results[r] for r in range(world_size)]
 
 

@@ -33,16 +33,17 @@ def _pipeline_gpu_worker(rank, world, tmpdir):
     nhalo = rp.recv_shape[peer]
     avg = 0.0
     for epoch in range(4):
+        # requires_grad from epoch 0 on: the grad hook must post a transfer
+        # EVERY epoch (1-epoch staleness), exactly like real training
         feat = torch.full((rp.num_in, F), float(epoch + 1 + 10 * rank),
-                          device="cuda:0", requires_grad=(epoch > 0))
+                          device="cuda:0", requires_grad=True)
         h = buf.update(1, feat)
         halo = h[rp.num_in:rp.num_in + nhalo]
         assert torch.allclose(halo.detach().cpu(),
                               torch.full((nhalo, F), avg), atol=1e-5), \
             f"epoch {epoch}: {halo[0,0].item()} != {avg}"
-        if epoch > 0:
-            h.sum().backward()
-            assert torch.isfinite(feat.grad).all()
+        h.sum().backward()
+        assert torch.isfinite(feat.grad).all()
         avg = 0.5 * avg + 0.5 * float(epoch + 1 + 10 * peer)
         buf.next_epoch()
         comm_timer.clear()
@@ -52,7 +53,7 @@ def _pipeline_gpu_worker(rank, world, tmpdir):
 
 
 def test_gpu_pipeline_two_ranks_one_device(tmp_path):
-    run_distributed(_pipeline_gpu_worker, WORLD, args=(str(tmp_path),))
+    run_distributed(_pipeline_gpu_worker, WORLD, args=(str(tmp_path),), timeout=900)
 
 
 def _train_gpu_worker(rank, world, tmpdir):
@@ -78,4 +79,4 @@ def _train_gpu_worker(rank, world, tmpdir):
 
 
 def test_gpu_two_rank_training_one_device(tmp_path):
-    run_distributed(_train_gpu_worker, WORLD, args=(str(tmp_path),))
+    run_distributed(_train_gpu_worker, WORLD, args=(str(tmp_path),), timeout=900)
