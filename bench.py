@@ -54,9 +54,10 @@ def main():
     backend = args.backend or ("nccl" if device.startswith("cuda")
                                else "gloo")
     if device.startswith("cuda"):
-        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-        torch.cuda.set_device(local_rank)
-        device = f"cuda:{local_rank}"
+        if args.device is None:
+            local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+            device = f"cuda:{local_rank}"
+        torch.cuda.set_device(torch.device(device))
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
