@@ -100,10 +100,36 @@ def inductive_split(u, v, n, ndata):
     return tr, va, (u, v, n, ndata)
 
 
+def exchange_halo_values(rp: RuntimePartition,
+                         values: torch.Tensor) -> torch.Tensor:
+    """One-shot fetch of a per-node scalar (e.g. in-degree) for halo nodes;
+    returns values extended to [num_all] in halo-slot order."""
+    size = dist.get_world_size() if dist.is_initialized() else 1
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    if size == 1:
+        return values
+    from pipegcn_amd.parallel.transport import RingTransport
+
+    send = [None] * size
+    recv = [None] * size
+    col = values.unsqueeze(1)
+    for j in range(size):
+        if j == rank:
+            continue
+        send[j] = col[rp.boundary[j]].contiguous()
+        recv[j] = torch.zeros(rp.recv_shape[j], 1, device=values.device)
+    RingTransport().all_to_all(send, recv, key="halo_scalar", tag=2)
+    return torch.cat([values] + [recv[j].squeeze(1) for j in range(size)
+                                 if j != rank])
+
+
 def precompute(rp: RuntimePartition, args) -> torch.Tensor:
     """--use-pp: one-shot raw-feature halo exchange + one mean-agg SpMM;
     returns [feat ‖ mean_feat] so layer 0 needs no per-epoch communication
     (reference /root/reference/train.py:169-189)."""
+    if args.model != "graphsage":
+        raise NotImplementedError(
+            "--use-pp supports graphsage only (reference parity)")
     feat = rp.ndata["feat"]
     size = dist.get_world_size() if dist.is_initialized() else 1
     rank = dist.get_rank() if dist.is_initialized() else 0
@@ -133,6 +159,12 @@ def create_model(layer_size, args):
         return GraphSAGE(layer_size, F.relu, args.use_pp, norm=args.norm,
                          dropout=args.dropout, n_linear=args.n_linear,
                          train_size=args.n_train)
+    if args.model == "gcn":
+        from pipegcn_amd.models.gcn import GCN
+
+        return GCN(layer_size, F.relu, args.use_pp, norm=args.norm,
+                   dropout=args.dropout, n_linear=args.n_linear,
+                   train_size=args.n_train)
     raise NotImplementedError(f"unknown model {args.model}")
 
 
@@ -183,6 +215,9 @@ def run(part: PartData, args, device: str = "cpu",
         feat = precompute(rp, args)
 
     in_deg = rp.ndata["in_degree"]
+    if args.model == "gcn":
+        # GCN's symmetric normalization needs halo-node degrees too
+        in_deg = exchange_halo_values(rp, in_deg)
     labels_all = rp.ndata["label"]
     num_train = rp.num_train
     part_train = max(num_train, 1)

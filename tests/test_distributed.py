@@ -260,7 +260,7 @@ def test_sync_batchnorm():
 # -------------------------------------------------- training equivalence
 
 
-def _equiv_worker(rank, world, tmpdir, pipeline):
+def _equiv_worker(rank, world, tmpdir, pipeline, model="graphsage"):
     from pipegcn_amd import trainer
     from pipegcn_amd.graph import partition
     from pipegcn_amd.parallel import context as ctx
@@ -273,7 +273,7 @@ def _equiv_worker(rank, world, tmpdir, pipeline):
     (u, v, n, ndata), part = _prepare_partitions(
         os.path.join(tmpdir, f"p{world}"), world)
     args = make_args(n_partitions=world,
-                     enable_pipeline=pipeline)
+                     enable_pipeline=pipeline, model=model)
     from pipegcn_amd.graph.datasets import data_stats
     args.n_feat, args.n_class, args.n_train = data_stats(ndata)
     s = trainer.run(part, args, device="cpu")
@@ -304,3 +304,15 @@ def test_pipelined_training_converges(tmp_path):
         functools.partial(_equiv_worker, pipeline=True), 2,
         args=(str(tmp_path),))[0]
     assert losses[-1] < losses[0]
+
+
+def _gcn_worker(rank, world, tmpdir):
+    return _equiv_worker(rank, world, tmpdir, False, model="gcn")
+
+
+def test_gcn_2part_matches_1part(tmp_path):
+    two = run_distributed(_gcn_worker, 2, args=(str(tmp_path),))[0]
+    one = run_distributed(_gcn_worker, 1, args=(str(tmp_path),))[0]
+    for a, b in zip(one, two):
+        assert abs(a - b) / max(abs(a), 1e-9) < 2e-3, (one, two)
+    assert two[-1] < two[0]
