@@ -12,10 +12,9 @@
 // Uses v_mfma_f32_32x32x2_f32 (f32 in / f32 accumulate — bitwise an fmaf
 // chain, NO TF32 anywhere; see cdna_hip_programming.md §3): 155 TF chip peak.
 // Structure: 128x128 block tile, 4 waves (2x2), each wave a 64x64 tile of
-// 2x2 32x32 fragments; K staged through LDS in 32-deep tiles, +1-dword row
-// padding for conflict-free ds_reads. Correctness-first single-buffer
-// staging — the kernel is MFMA-issue-bound at fp32 rate, so deep staging
-// pipelines buy little here (unlike bf16).
+// 2x2 32x32 fragments; K staged through LDS in TBK-deep tiles, +1-dword row
+// padding for conflict-free ds_reads; interior blocks (the vast majority at
+// M ~ 10^5..10^6 rows) take a guard-free staging fast path.
 //
 // Weights are in torch Linear layout [N, K] row-major; they are tiny
 // (<= 616 KB) and L2-resident, so the transposed LDS fill reads them
@@ -26,22 +25,24 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
 namespace {
 
 using f32x16 = __attribute__((__vector_size__(16 * sizeof(float)))) float;
 
 constexpr int BM = 128;
 constexpr int BN = 128;
-constexpr int BK = 32;
 constexpr int PAD = 1;
 
+template <int TBK>
 __global__ __launch_bounds__(256) void sage_dual_gemm_kernel(
     const float* __restrict__ x1, const float* __restrict__ x2,
     const float* __restrict__ w1, const float* __restrict__ w2,
     const float* __restrict__ bias, float* __restrict__ out, int64_t M,
     int64_t N, int64_t K) {
-  __shared__ float a_lds[BM][BK + PAD];
-  __shared__ float b_lds[BK][BN + 4];
+  __shared__ float a_lds[BM][TBK + PAD];
+  __shared__ float b_lds[TBK][BN + 4];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -51,74 +52,84 @@ __global__ __launch_bounds__(256) void sage_dual_gemm_kernel(
 
   const int64_t m0 = static_cast<int64_t>(blockIdx.x) * BM;
   const int64_t n0 = static_cast<int64_t>(blockIdx.y) * BN;
+  const bool interior = (m0 + BM <= M) && (n0 + BN <= N);
 
   f32x16 acc[2][2] = {};
 
   const int l31 = lane & 31;
   const int lk = lane >> 5;  // 0/1 : k within the MFMA K=2 step
 
+  // staging geometry: A: thread covers (tid>>per) rows x elems
+  const int a_r = tid / (TBK / 4);          // A row per float4 (TBK/4 thr/row)
+  const int a_c = (tid % (TBK / 4)) * 4;    // col within tile
+  constexpr int A_RSTEP = 256 / (TBK / 4);  // rows covered per pass
+  const int b_n = tid >> 1;                 // 0..127 (W row)
+  const int b_k = (tid & 1) * (TBK / 2);    // k base
+
   for (int seg = 0; seg < 2; ++seg) {
     const float* X = seg == 0 ? x1 : x2;
     const float* W = seg == 0 ? w1 : w2;
-    for (int64_t k0 = 0; k0 < K; k0 += BK) {
-      // ---- stage A tile [BM][BK]: 256 threads x 16 elem = 4 x float4.
-      // thread t covers row = 8 rows/iter pattern: t/8 row, (t%8)*4 col
-      {
-        const int ar = tid >> 3;          // 0..31
-        const int ac = (tid & 7) * 4;     // 0..28
+    for (int64_t k0 = 0; k0 < K; k0 += TBK) {
+      const bool kfull = (k0 + TBK <= K);
+      // ---- stage A tile [BM][TBK]
+      if (interior && kfull) {
 #pragma unroll
-        for (int rr = 0; rr < 4; ++rr) {
-          const int row = ar + rr * 32;
+        for (int rr = 0; rr < BM / A_RSTEP; ++rr) {
+          const int row = a_r + rr * A_RSTEP;
+          const float4 val = *reinterpret_cast<const float4*>(
+              X + (m0 + row) * K + k0 + a_c);
+          a_lds[row][a_c + 0] = val.x;
+          a_lds[row][a_c + 1] = val.y;
+          a_lds[row][a_c + 2] = val.z;
+          a_lds[row][a_c + 3] = val.w;
+        }
+      } else {
+#pragma unroll
+        for (int rr = 0; rr < BM / A_RSTEP; ++rr) {
+          const int row = a_r + rr * A_RSTEP;
           const int64_t gm = m0 + row;
-          float4 val = {0.f, 0.f, 0.f, 0.f};
+          float tmp[4] = {0.f, 0.f, 0.f, 0.f};
           if (gm < M) {
-            const int64_t gk = k0 + ac;
-            if (gk + 3 < K) {
-              val = *reinterpret_cast<const float4*>(X + gm * K + gk);
-            } else {
-              float tmp[4] = {0.f, 0.f, 0.f, 0.f};
-              for (int q = 0; q < 4 && gk + q < K; ++q)
-                tmp[q] = X[gm * K + gk + q];
-              val = {tmp[0], tmp[1], tmp[2], tmp[3]};
-            }
+            for (int q = 0; q < 4 && k0 + a_c + q < K; ++q)
+              tmp[q] = X[gm * K + k0 + a_c + q];
           }
-          a_lds[row][ac + 0] = val.x;
-          a_lds[row][ac + 1] = val.y;
-          a_lds[row][ac + 2] = val.z;
-          a_lds[row][ac + 3] = val.w;
+          a_lds[row][a_c + 0] = tmp[0];
+          a_lds[row][a_c + 1] = tmp[1];
+          a_lds[row][a_c + 2] = tmp[2];
+          a_lds[row][a_c + 3] = tmp[3];
         }
       }
-      // ---- stage B tile [BK][BN] = W[n0:n0+BN][k0:k0+BK] transposed.
-      // 256 threads cover 128 n-rows x 32 k-cols: t/2 = n, (t%2)*16 = k base
-      {
-        const int bn = tid >> 1;           // 0..127
-        const int bk = (tid & 1) * 16;     // 0 or 16
-        const int64_t gn = n0 + bn;
+      // ---- stage B tile [TBK][BN] = W[n0:n0+BN][k0:k0+TBK] transposed
+      if (interior && kfull) {
 #pragma unroll
-        for (int q = 0; q < 16; q += 4) {
-          const int64_t gk = k0 + bk + q;
-          float4 val = {0.f, 0.f, 0.f, 0.f};
+        for (int q = 0; q < TBK / 2; q += 4) {
+          const float4 val = *reinterpret_cast<const float4*>(
+              W + (n0 + b_n) * K + k0 + b_k + q);
+          b_lds[b_k + q + 0][b_n] = val.x;
+          b_lds[b_k + q + 1][b_n] = val.y;
+          b_lds[b_k + q + 2][b_n] = val.z;
+          b_lds[b_k + q + 3][b_n] = val.w;
+        }
+      } else {
+        const int64_t gn = n0 + b_n;
+#pragma unroll
+        for (int q = 0; q < TBK / 2; q += 4) {
+          float tmp[4] = {0.f, 0.f, 0.f, 0.f};
           if (gn < N) {
-            if (gk + 3 < K) {
-              val = *reinterpret_cast<const float4*>(W + gn * K + gk);
-            } else {
-              float tmp[4] = {0.f, 0.f, 0.f, 0.f};
-              for (int p = 0; p < 4 && gk + p < K; ++p)
-                tmp[p] = W[gn * K + gk + p];
-              val = {tmp[0], tmp[1], tmp[2], tmp[3]};
-            }
+            for (int p = 0; p < 4 && k0 + b_k + q + p < K; ++p)
+              tmp[p] = W[gn * K + k0 + b_k + q + p];
           }
-          b_lds[bk + q + 0][bn] = val.x;
-          b_lds[bk + q + 1][bn] = val.y;
-          b_lds[bk + q + 2][bn] = val.z;
-          b_lds[bk + q + 3][bn] = val.w;
+          b_lds[b_k + q + 0][b_n] = tmp[0];
+          b_lds[b_k + q + 1][b_n] = tmp[1];
+          b_lds[b_k + q + 2][b_n] = tmp[2];
+          b_lds[b_k + q + 3][b_n] = tmp[3];
         }
       }
       __syncthreads();
 
-      // ---- MFMA inner loop: 16 K-steps of 2
+      // ---- MFMA inner loop: TBK/2 K-steps of 2
 #pragma unroll
-      for (int kk = 0; kk < BK; kk += 2) {
+      for (int kk = 0; kk < TBK; kk += 2) {
         float a0 = a_lds[wr * 64 + l31][kk + lk];
         float a1 = a_lds[wr * 64 + 32 + l31][kk + lk];
         float b0 = b_lds[kk + lk][wc * 64 + l31];
@@ -177,10 +188,20 @@ void sage_dual_gemm_hip(torch::Tensor x1, torch::Tensor x2, torch::Tensor w1,
   }
   dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  hipLaunchKernelGGL(sage_dual_gemm_kernel, grid, dim3(256), 0, stream,
-                     x1.data_ptr<float>(), x2.data_ptr<float>(),
-                     w1.data_ptr<float>(), w2.data_ptr<float>(), bp,
-                     out.data_ptr<float>(), M, N, K);
+  int bk = K > 512 ? 64 : 32;  // measured: BK=64 wins at K=602, 32 at 256
+  if (const char* e = std::getenv("PIPEGCN_GEMM_BK")) bk = std::atoi(e);
+  if (bk == 32)
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(sage_dual_gemm_kernel<32>), grid,
+                       dim3(256), 0, stream, x1.data_ptr<float>(),
+                       x2.data_ptr<float>(), w1.data_ptr<float>(),
+                       w2.data_ptr<float>(), bp, out.data_ptr<float>(), M, N,
+                       K);
+  else
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(sage_dual_gemm_kernel<64>), grid,
+                       dim3(256), 0, stream, x1.data_ptr<float>(),
+                       x2.data_ptr<float>(), w1.data_ptr<float>(),
+                       w2.data_ptr<float>(), bp, out.data_ptr<float>(), M, N,
+                       K);
   hipError_t e = hipGetLastError();
   TORCH_CHECK(e == hipSuccess, "HIP error: ", hipGetErrorString(e));
 }
