@@ -37,6 +37,13 @@ def main():
     ap.add_argument("--n-layers", type=int, default=3)
     ap.add_argument("--no-pipeline", action="store_true",
                     help="disable cross-epoch pipelining (vanilla mode)")
+    ap.add_argument("--use-pp", action="store_true",
+                    help="precompute layer-0 aggregation (reference "
+                         "headline config)")
+    ap.add_argument("--norm", type=str, default="layer",
+                    choices=["layer", "batch", "none"])
+    ap.add_argument("--feat-corr", action="store_true")
+    ap.add_argument("--grad-corr", action="store_true")
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--backend", type=str, default=None,
                     help="override torch.distributed backend (tests)")
@@ -67,32 +74,47 @@ def main():
     from pipegcn_amd.graph.synthetic import SHAPES, synth_partition
     from pipegcn_amd.models.sage import GraphSAGE
     from pipegcn_amd.parallel import context as ctx
-    from pipegcn_amd.trainer import get_layer_size
+    from pipegcn_amd.trainer import get_layer_size, precompute
     from pipegcn_amd.utils.timer import comm_timer
     import torch.nn.functional as F
+    import types
 
     torch.manual_seed(0)
-    part = synth_partition(args.shape, rank, world, seed=0)
+    # batch norm normalizes by the GLOBAL train count (reference semantics,
+    # consistent only when every node is a train node — the inductive setup)
+    train_frac = 1.0 if args.norm == "batch" else 0.66
+    part = synth_partition(args.shape, rank, world, seed=0,
+                           train_frac=train_frac)
     rp = build_runtime_partition(part, device=device)
 
     pipeline = not args.no_pipeline
+    norm = None if args.norm == "none" else args.norm
+    multilabel = SHAPES[args.shape][4]
     layer_size = get_layer_size(part.n_feat, args.n_hidden, part.n_class,
                                 args.n_layers)
     comm_group = dist.new_group(backend=backend) if world > 1 else None
     ctx.buffer.init_buffer(rp.num_in, rp.num_all, rp.boundary, rp.recv_shape,
-                           layer_size[: args.n_layers], use_pp=False,
+                           layer_size[: args.n_layers], use_pp=args.use_pp,
                            backend=backend, pipeline=pipeline,
+                           corr_feat=args.feat_corr,
+                           corr_grad=args.grad_corr,
                            device=device, group=comm_group,
                            collect_stats=True)
 
-    model = GraphSAGE(layer_size, F.relu, use_pp=False, dropout=0.5,
-                      norm="layer", n_linear=0,
+    model = GraphSAGE(layer_size, F.relu, use_pp=args.use_pp, dropout=0.5,
+                      norm=norm, n_linear=0,
                       train_size=part.n_train).to(device)
     ctx.reducer.init(model)
-    loss_fcn = torch.nn.CrossEntropyLoss(reduction="sum")
+    if multilabel:
+        loss_fcn = torch.nn.BCEWithLogitsLoss(reduction="sum")
+    else:
+        loss_fcn = torch.nn.CrossEntropyLoss(reduction="sum")
     optimizer = torch.optim.Adam(model.parameters(), lr=0.01)
 
     feat = rp.ndata["feat"]
+    if args.use_pp:
+        pp_args = types.SimpleNamespace(model="graphsage")
+        feat = precompute(rp, pp_args)
     in_deg = rp.ndata["in_degree"]
     labels = rp.ndata["label"][: rp.num_train]
     model.train()
@@ -169,6 +191,8 @@ def main():
                 "n_feat": n_feat,
                 "n_class": n_class,
                 "pipeline": pipeline,
+                "use_pp": args.use_pp,
+                "norm": args.norm,
                 "parallelism": f"graph-partition dp{world}",
                 "comm_busy_s_per_epoch": comm_busy / args.steps,
                 "comm_wait_s_per_epoch": wait_s / args.steps,
