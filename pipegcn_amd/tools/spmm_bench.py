@@ -28,13 +28,15 @@ def build(n=232_965, avg_deg=492, seed=0):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--f", type=int, nargs="+", default=[602, 256])
+    ap.add_argument("--nodes", type=int, default=232_965)
+    ap.add_argument("--deg", type=int, default=492)
     ap.add_argument("--rounds", type=int, default=5)
     ap.add_argument("--iters", type=int, default=3)
     args = ap.parse_args()
 
     from pipegcn_amd import ops
 
-    hg, E = build()
+    hg, E = build(args.nodes, args.deg)
     n = hg.num_in
     deg = hg.csr.row_degrees().to("cuda").clamp(min=1)
     inv = (1.0 / deg).contiguous()
