@@ -252,27 +252,30 @@ __global__ void ema_kernel(float* __restrict__ avg,
   }
 }
 
-int pick_vec(int64_t F, int64_t num_src) {
+int pick_vec(int64_t F, int64_t num_rows) {
   // Prefer the widest load whose column panel (num_src × 64·VEC × 4 B) stays
   // L3-resident (256 MiB Infinity Cache; target ≤ ~160 MiB leaves room for
   // the streaming index array). Overridable for A/B benchmarking.
-  // Measured on MI355X (profiles/spmm_ab.md): the gather path tops out at
-  // ~6.5-7 TB/s logical regardless of L3 panel residency; what matters is
-  // pair-level load balance — prefer the VEC giving ~3-8 column chunks,
-  // else the widest aligned load. (num_src kept for future blocking.)
-  (void)num_src;
+  // Measured on MI355X (profiles/README.md): the gather path tops out at
+  // ~6.5-7 TB/s logical regardless of L3 panel residency. What matters is
+  // (row,chunk)-pair count: with heavy-tailed degrees, few pairs leave the
+  // tail imbalanced (Reddit 233k rows: VEC1 with 4x the pairs beats VEC4 by
+  // 8%), while with plenty of pairs the widest load wins (ogbn-products
+  // 2.45M rows: VEC4 beats VEC1 by 17%). Rule: widest VEC whose pair count
+  // reaches ~40 per resident wave (16384 waves -> 640k pairs), else widest.
   if (const char* e = std::getenv("PIPEGCN_SPMM_VEC")) {
     int v = std::atoi(e);
     if ((v == 4 || v == 2 || v == 1) && F % v == 0) return v;
   }
-  int best_any = 0;
+  constexpr int64_t kTargetPairs = 640 * 1024;
+  int widest = 0;
   for (int v : {4, 2, 1}) {
     if (F % v != 0) continue;
-    if (!best_any) best_any = v;
+    if (!widest) widest = v;
     const int64_t nchunks = (F + 64 * v - 1) / (64 * v);
-    if (nchunks >= 3 && nchunks <= 8) return v;
+    if (num_rows * nchunks >= kTargetPairs) return v;
   }
-  return best_any;
+  return widest;
 }
 
 }  // namespace
@@ -300,7 +303,7 @@ void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
     ssp = src_scale.data_ptr<float>();
   }
   auto stream = current_stream();
-  const int vec = pick_vec(F, num_src);
+  const int vec = pick_vec(F, num_rows);
   bool chunk_outer = false;  // measured: chunk-inner wins at every shape
   if (const char* e = std::getenv("PIPEGCN_SPMM_ORDER"))
     chunk_outer = (e[0] == 'o');
