@@ -346,4 +346,7 @@ def init_processes(rank, size, args, device: Optional[str] = None):
         torch.cuda.set_device(torch.device(device))
     graph_dir = os.path.join("partitions", args.graph_name)
     part = load_partition(graph_dir, rank)
-    return run(part, args, device=device)
+    try:
+        return run(part, args, device=device)
+    finally:
+        dist.destroy_process_group()
