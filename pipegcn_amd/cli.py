@@ -71,6 +71,15 @@ def create_parser(argv=None):
     parser.add_argument("--parts-per-node", "--parts_per_node", type=int,
                         default=10)
 
+    parser.add_argument("--checkpoint-every", "--checkpoint_every",
+                        type=int, default=0,
+                        help="save a per-rank training checkpoint every N "
+                             "epochs under checkpoint/ (0 = off; extension "
+                             "over the reference, which created checkpoint/ "
+                             "but never used it)")
+    parser.add_argument("--resume", action="store_true",
+                        help="resume from the latest checkpoint/ state")
+
     parser.add_argument("--eval", action="store_true",
                         help="enable evaluation")
     parser.add_argument("--no-eval", action="store_false", dest="eval",

@@ -254,7 +254,21 @@ def run(part: PartData, args, device: str = "cpu",
     if device != "cpu":
         torch.cuda.reset_peak_memory_stats()
 
-    for epoch in range(args.n_epochs):
+    # --- checkpoint/resume (extension; reference has save-only final model)
+    ckpt_every = getattr(args, "checkpoint_every", 0)
+    ckpt_path = os.path.join(
+        "checkpoint", f"{args.graph_name}_rank{rank}.ckpt")
+    start_epoch = 0
+    if getattr(args, "resume", False) and os.path.exists(ckpt_path):
+        state = torch.load(ckpt_path, map_location=device,
+                           weights_only=True)
+        model.load_state_dict(state["model"])
+        optimizer.load_state_dict(state["optimizer"])
+        start_epoch = state["epoch"] + 1
+        best_acc = state.get("best_acc", 0.0)
+        print(f"Process {rank:03d} | resumed from epoch {state['epoch']}")
+
+    for epoch in range(start_epoch, args.n_epochs):
         t0 = time.time()
         model.train()
         logits = model(rp.graph, feat, in_deg)
@@ -287,6 +301,12 @@ def run(part: PartData, args, device: str = "cpu",
         comm_timer.clear()
         loss_hist.append(loss.item())
         del loss
+
+        if ckpt_every > 0 and (epoch + 1) % ckpt_every == 0:
+            os.makedirs("checkpoint/", exist_ok=True)
+            torch.save({"epoch": epoch, "model": model.state_dict(),
+                        "optimizer": optimizer.state_dict(),
+                        "best_acc": best_acc}, ckpt_path)
 
         if (rank == 0 and args.eval
                 and (epoch + 1) % args.log_every == 0):

@@ -48,3 +48,16 @@ def test_main_random_partition_use_pp(tmp_path):
     out = run_main(tmp_path, ["--partition-method", "random", "--use-pp",
                               "--no-eval"])
     assert "Epoch" in out
+
+
+def test_main_gcn_model(tmp_path):
+    out = run_main(tmp_path, ["--model", "gcn", "--no-eval",
+                              "--enable-pipeline"])
+    assert "Epoch" in out
+
+
+def test_checkpoint_resume(tmp_path):
+    run_main(tmp_path, ["--no-eval", "--checkpoint-every", "4"])
+    out = run_main(tmp_path, ["--no-eval", "--checkpoint-every", "4",
+                              "--resume"])
+    assert "resumed from epoch 7" in out
