@@ -316,3 +316,68 @@ def test_gcn_2part_matches_1part(tmp_path):
     for a, b in zip(one, two):
         assert abs(a - b) / max(abs(a), 1e-9) < 2e-3, (one, two)
     assert two[-1] < two[0]
+
+
+# --------------------------------------------- disconnected partition pair
+
+
+def _chain_worker(rank, world, tmpdir):
+    """3 partitions in a chain 0-1-2: ranks 0 and 2 share NO edges — their
+    boundary/recv sets are empty and the batched P2P exchange must skip them
+    consistently on both sides (RCCL hazard at 8 GPUs)."""
+    from pipegcn_amd.graph import partition
+    from pipegcn_amd.graph.halo import build_runtime_partition
+    from pipegcn_amd.parallel.buffer import Buffer
+
+    if rank == 0:
+        # chain graph: 3 blocks of 10 nodes; edges only within a block or
+        # to the adjacent block; plus self-loops
+        u, v = [], []
+        n = 30
+        for i in range(n):
+            u.append(i)
+            v.append(i)
+            blk = i // 10
+            for j in range(3):
+                t = (i + j * 3 + 1) % 10 + blk * 10
+                u.append(t)
+                v.append(i)
+            if blk < 2:  # edge from next block
+                u.append(i + 10)
+                v.append(i)
+        # force block partition by pre-assigning: use random method w/ seed
+        # -> instead use explicit partition via contiguous split: the
+        # partitioner with balanced blocks on a chain finds blocks anyway
+        partition.partition_and_save(torch.tensor(u), torch.tensor(v), n,
+                                     {"feat": torch.randn(n, 4),
+                                      "label": torch.randint(0, 3, (n,)),
+                                      "train_mask": torch.ones(n).bool(),
+                                      "val_mask": torch.zeros(n).bool(),
+                                      "test_mask": torch.zeros(n).bool()},
+                                     tmpdir, 3, "metis", "cut", 1)
+    dist.barrier()
+    part = partition.load_partition(tmpdir, rank)
+    rp = build_runtime_partition(part)
+    # sanity: SOME pair must be empty for the test to mean anything;
+    # find it dynamically (partitioner may order blocks arbitrarily)
+    empties = sum(1 for j in range(world)
+                  if j != rank and rp.recv_shape[j] == 0)
+    buf = Buffer()
+    buf.init_buffer(rp.num_in, rp.num_all, rp.boundary, rp.recv_shape,
+                    [4, 4], pipeline=True, backend="gloo")
+    for epoch in range(3):
+        feat = torch.full((rp.num_in, 4), float(epoch + 1),
+                          requires_grad=True)
+        h = buf.update(1, feat)
+        h.sum().backward()
+        buf.next_epoch()
+        comm_timer.clear()
+    buf.synchronize()
+    buf.shutdown()
+    return empties
+
+
+def test_chain_partition_empty_boundary(tmp_path):
+    empties = run_distributed(_chain_worker, 3, args=(str(tmp_path),))
+    # the chain's two end partitions each see one empty peer
+    assert sum(empties) >= 2, empties
