@@ -25,8 +25,11 @@ class SyncBatchNormFunc(Function):
         if not training:
             mean, var = running_mean, running_var
         else:
-            sum_x = x.sum(dim=0)
-            sum_x2 = (x * x).sum(dim=0)
+            # GEMV instead of eager .sum(0): ~20x on tall [N,F] (ROCm eager
+            # column reduction is slow — see profiles/README.md)
+            ones = x.new_ones(x.size(0))
+            sum_x = torch.mv(x.t(), ones)
+            sum_x2 = torch.mv((x * x).t(), ones)
             _maybe_all_reduce(sum_x)
             _maybe_all_reduce(sum_x2)
             mean = sum_x / whole_size
@@ -43,8 +46,9 @@ class SyncBatchNormFunc(Function):
     @staticmethod
     def backward(ctx, grad):
         x_hat, weight, std = ctx.saved_tensors
-        dbias = grad.sum(dim=0)
-        dweight = (grad * x_hat).sum(dim=0)
+        ones = grad.new_ones(grad.size(0))
+        dbias = torch.mv(grad.t(), ones)
+        dweight = torch.mv((grad * x_hat).t(), ones)
         _maybe_all_reduce(dbias)
         _maybe_all_reduce(dweight)
         n = ctx.whole_size
