@@ -60,6 +60,11 @@ static torch::Tensor sage_dual_gemm(torch::Tensor x1, torch::Tensor x2,
   return out;
 }
 
+static torch::Tensor colsum(torch::Tensor x) {
+  if (x.is_cuda()) return colsum_hip(x);
+  return x.sum(0);
+}
+
 static void ema_update(torch::Tensor avg, torch::Tensor x, double momentum) {
   if (avg.is_cuda()) {
     ema_update_hip(avg, x, momentum);
@@ -79,6 +84,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "gather into a preallocated out buffer");
   m.def("scatter_add_rows", &scatter_add_rows, "dst[idx[i],:] += src[i,:]");
   m.def("ema_update", &ema_update, "avg = m*avg + (1-m)*x");
+  m.def("colsum", &colsum, "out[n] = sum_m x[m,n]");
   m.def("sage_dual_gemm", &sage_dual_gemm,
         "out = x1 @ w1^T + x2 @ w2^T + bias (MFMA fp32, fused)");
   m.attr("with_hip") = true;

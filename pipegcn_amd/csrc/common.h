@@ -63,6 +63,9 @@ void scatter_add_rows_hip(torch::Tensor dst, torch::Tensor idx,
 // avg = momentum * avg + (1 - momentum) * x
 void ema_update_hip(torch::Tensor avg, torch::Tensor x, double momentum);
 
+// out[n] = sum_m x[m, n] — fast column reduction for bias/BN grads.
+torch::Tensor colsum_hip(torch::Tensor x);
+
 // Fused dual GEMM for the GraphSAGE layer epilogue:
 //   out[M,N] = x1[M,K] @ w1t[K,N] + x2[M,K] @ w2t[K,N] + b[N]
 // fp32, MFMA (v_mfma_f32_16x16x4_f32). Weights pre-transposed to [K,N].

@@ -13,6 +13,12 @@ from torch import nn
 from torch.autograd import Function
 
 
+def _colsum(x):
+    from pipegcn_amd import native
+
+    return native().colsum(x.contiguous())
+
+
 def _maybe_all_reduce(t):
     if dist.is_initialized() and dist.get_world_size() > 1:
         dist.all_reduce(t, op=dist.ReduceOp.SUM)
@@ -25,11 +31,8 @@ class SyncBatchNormFunc(Function):
         if not training:
             mean, var = running_mean, running_var
         else:
-            # GEMV instead of eager .sum(0): ~20x on tall [N,F] (ROCm eager
-            # column reduction is slow — see profiles/README.md)
-            ones = x.new_ones(x.size(0))
-            sum_x = torch.mv(x.t(), ones)
-            sum_x2 = torch.mv((x * x).t(), ones)
+            sum_x = _colsum(x)
+            sum_x2 = _colsum(x * x)
             _maybe_all_reduce(sum_x)
             _maybe_all_reduce(sum_x2)
             mean = sum_x / whole_size
@@ -46,9 +49,8 @@ class SyncBatchNormFunc(Function):
     @staticmethod
     def backward(ctx, grad):
         x_hat, weight, std = ctx.saved_tensors
-        ones = grad.new_ones(grad.size(0))
-        dbias = torch.mv(grad.t(), ones)
-        dweight = torch.mv((grad * x_hat).t(), ones)
+        dbias = _colsum(grad)
+        dweight = _colsum(grad * x_hat)
         _maybe_all_reduce(dbias)
         _maybe_all_reduce(dweight)
         n = ctx.whole_size

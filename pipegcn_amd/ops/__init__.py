@@ -106,9 +106,7 @@ class _SageDualLinear(torch.autograd.Function):
         gx2 = g @ w2
         gw1 = g.t() @ x1
         gw2 = g.t() @ x2
-        # bias grad as a GEMV: eager .sum(0) on ROCm runs at ~130 GB/s for
-        # tall [M,N] tensors (profiles/README.md) — 20x slower than rocBLAS
-        gb = torch.mv(g.t(), g.new_ones(g.size(0)))
+        gb = native().colsum(g)  # two-phase column sum (fastest measured)
         return gx1, gx2, gw1, gw2, gb, gb
 
 
