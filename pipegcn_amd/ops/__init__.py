@@ -93,6 +93,14 @@ class _SageDualLinear(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x1, x2, w1, w2, b1, b2):
         ctx.save_for_backward(x1, x2, w1, w2)
+        if w1.size(0) < 64 or not x1.is_cuda:
+            # thin-N / CPU: the 128x128 MFMA tile would waste most of the
+            # block; rocBLAS pair instead (the op is memory-bound there)
+            out = torch.mm(x1, w1.t())
+            out.addmm_(x2, w2.t())
+            if b1 is not None:
+                out.add_(b1 + b2)
+            return out
         bias = (b1 + b2) if b1 is not None else torch.Tensor()
         return native().sage_dual_gemm(x1.contiguous(), x2.contiguous(),
                                        w1.contiguous(), w2.contiguous(),
@@ -113,11 +121,10 @@ class _SageDualLinear(torch.autograd.Function):
 def sage_dual_linear(x1, x2, lin1, lin2):
     """Fused linear1(x1) + linear2(x2) for the SAGE layer.
 
-    The MFMA kernel's 128x128 tile wastes compute on thin outputs (final
-    layer, N = n_class < 64) where the op is memory-bound anyway — those go
-    to rocBLAS (measured: 0.28 vs 0.49 ms at N=41).
+    All shapes go through _SageDualLinear so the bias grad uses the native
+    colsum — torch's nn.Linear backward picks a pathological 1024-thread
+    reduce for thin odd N ([2.45M,47] bias grad measured 18.7 ms vs 0.2).
+    The forward dispatches thin-N / CPU to rocBLAS internally.
     """
-    if lin1.weight.size(0) < 64:
-        return lin1(x1) + lin2(x2)
     return _SageDualLinear.apply(x1, x2, lin1.weight, lin2.weight, lin1.bias,
                                  lin2.bias)
