@@ -34,6 +34,7 @@ Semantics preserved exactly (checked against the reference):
 """
 from __future__ import annotations
 
+import os
 import queue
 import threading
 import time
@@ -41,6 +42,11 @@ from typing import List, Optional
 
 import torch
 import torch.distributed as dist
+
+# PIPEGCN_DEBUG=1: log a checksum of every transfer's send/recv payload
+# (per epoch/layer/direction) — the race-hunting facility the reference
+# lacks (SURVEY §5 race detection).
+_DEBUG = os.environ.get("PIPEGCN_DEBUG", "0") == "1"
 
 from pipegcn_amd import ops
 from pipegcn_amd.parallel.transport import RingTransport
@@ -347,6 +353,17 @@ class Buffer:
                                  self._f_send[layer][j])
         self._transport.all_to_all(self._f_send[layer], self._f_recv[layer],
                                    key=("f", layer), tag=tag)
+        if _DEBUG:
+            self._debug_log("feat", tag, self._f_send[layer],
+                            self._f_recv[layer])
+
+    def _debug_log(self, kind, tag, send, recv):
+        def csum(ts):
+            return [round(float(t.double().sum().item()), 4)
+                    for t in ts if t is not None]
+
+        print(f"[pipegcn-debug r{self._rank}] {kind} tag={tag} "
+              f"send={csum(send)} recv={csum(recv)}", flush=True)
 
     def _do_grad_transfer(self, epoch, layer, grad, ready):
         tag = epoch * 2 * self._n_layers + layer + self._n_layers
@@ -361,6 +378,8 @@ class Buffer:
             with torch.cuda.stream(self._comm_stream):
                 self._transport.all_to_all(send, self._b_recv[layer],
                                            key=("b", layer), tag=tag)
+                if _DEBUG:
+                    self._debug_log("grad", tag, send, self._b_recv[layer])
             done_stream = self._comm_stream
             if self._corr_grad:
                 self._corr_stream.wait_stream(self._comm_stream)
@@ -377,6 +396,8 @@ class Buffer:
             t0 = self._stats_begin()
             self._transport.all_to_all(send, self._b_recv[layer],
                                        key=("b", layer), tag=tag)
+            if _DEBUG:
+                self._debug_log("grad", tag, send, self._b_recv[layer])
             if self._corr_grad:
                 for j in range(self._size):
                     if j != self._rank:
