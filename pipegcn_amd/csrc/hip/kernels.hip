@@ -32,6 +32,8 @@ namespace {
 
 constexpr int kWave = 64;
 
+using int32x4 = __attribute__((ext_vector_type(4))) int;
+
 inline hipStream_t current_stream() {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 }
@@ -85,14 +87,23 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
     int64_t e = e_begin;
 
     if (full) {
-      // main path: unroll 4 edges so 4 gathers are in flight per lane.
-      // Index loads and output stores are NON-TEMPORAL: they are streaming
-      // traffic and must not evict the L3-resident feature panel.
+      // peel until e is 4-aligned so the unrolled loop can read FOUR int32
+      // indices with ONE dwordx4 — per 4 edges the VMEM stream is then
+      // 1 index load + 4 row gathers instead of 8 instructions (the gather
+      // path is instruction-rate-limited at small F; profiles/README.md).
+      for (; e < e_end && (e & 3); ++e) {
+        const int64_t u = indices[e];
+        const float s = HAS_SRC_SCALE ? src_scale[u] : 1.f;
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) acc[k] += s * feat[u * F + f0 + k];
+      }
       for (; e + 4 <= e_end; e += 4) {
-        const int64_t u0 = __builtin_nontemporal_load(indices + e + 0);
-        const int64_t u1 = __builtin_nontemporal_load(indices + e + 1);
-        const int64_t u2 = __builtin_nontemporal_load(indices + e + 2);
-        const int64_t u3 = __builtin_nontemporal_load(indices + e + 3);
+        const int32x4 uu = __builtin_nontemporal_load(
+            reinterpret_cast<const int32x4*>(indices + e));
+        const int64_t u0 = uu[0];
+        const int64_t u1 = uu[1];
+        const int64_t u2 = uu[2];
+        const int64_t u3 = uu[3];
         float s0 = 1.f, s1 = 1.f, s2 = 1.f, s3 = 1.f;
         if (HAS_SRC_SCALE) {
           s0 = src_scale[u0];
