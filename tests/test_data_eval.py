@@ -114,3 +114,33 @@ def test_eval_matches_train_forward():
         logits_ev = model(fg, ndata["feat"])
     assert torch.allclose(logits_tr, logits_ev, atol=1e-5), \
         (logits_tr - logits_ev).abs().max()
+
+
+def test_yelp_reference_layout(tmp_path, monkeypatch):
+    """Load the reference's on-disk Yelp format (adj_full.npz + feats.npy +
+    class_map.json + role.json — /root/reference/helper/utils.py:33-71)."""
+    import scipy.sparse as sp
+
+    n, f, c = 40, 6, 4
+    rng = np.random.RandomState(0)
+    adj = sp.random(n, n, density=0.2, format="csr", random_state=rng)
+    d = tmp_path / "dataset" / "yelp"
+    os.makedirs(d)
+    sp.save_npz(d / "adj_full.npz", adj)
+    np.save(d / "feats.npy", rng.randn(n, f))
+    cm = {str(i): rng.randint(0, 2, c).tolist() for i in range(n)}
+    with open(d / "class_map.json", "w") as fh:
+        json.dump(cm, fh)
+    idx = rng.permutation(n)
+    role = {"tr": idx[:20].tolist(), "va": idx[20:30].tolist(),
+            "te": idx[30:].tolist()}
+    with open(d / "role.json", "w") as fh:
+        json.dump(role, fh)
+    monkeypatch.chdir(tmp_path)
+    u, v, nn_, ndata = datasets.load_data("yelp")
+    assert nn_ == n
+    assert ndata["label"].shape == (n, c)
+    assert ndata["train_mask"].sum() == 20
+    # features are StandardScaler-normalized over train rows
+    tr = ndata["feat"][ndata["train_mask"]]
+    assert tr.mean(0).abs().max() < 0.5
