@@ -92,175 +92,4 @@ torch::Tensor spmm_cpu(torch::Tensor indptr, torch::Tensor indices,
   return out;
 }
 
-// ---------------------------------------------------------------------------
-// Partitioner
-// ---------------------------------------------------------------------------
-
-torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
-                                  int64_t nparts, int64_t objective,
-                                  double balance_slack, int64_t n_refine_passes,
-                                  int64_t seed) {
-  TORCH_CHECK(nparts >= 1, "nparts must be >= 1");
-  const int64_t N = indptr.numel() - 1;
-  auto part = torch::full({N}, -1, torch::kInt);
-  int32_t* pp = part.data_ptr<int32_t>();
-  if (nparts == 1) {
-    part.zero_();
-    return part;
-  }
-  const int64_t* ip = indptr.data_ptr<int64_t>();
-  const int32_t* xp = indices.data_ptr<int32_t>();
-
-  std::mt19937_64 rng(seed);
-  const int64_t cap =
-      static_cast<int64_t>((double)N / nparts * (1.0 + balance_slack)) + 1;
-
-  // --- seed selection: first seed random, others by repeated farthest-point
-  // BFS so seeds land in distant regions of the graph.
-  std::vector<int64_t> seeds;
-  std::vector<int32_t> dist(N, -1);
-  {
-    std::uniform_int_distribution<int64_t> uni(0, N - 1);
-    int64_t s0 = uni(rng);
-    seeds.push_back(s0);
-    // multi-source BFS from current seeds; next seed = farthest node
-    for (int64_t k = 1; k < nparts; ++k) {
-      std::fill(dist.begin(), dist.end(), -1);
-      std::deque<int64_t> q;
-      for (int64_t s : seeds) {
-        dist[s] = 0;
-        q.push_back(s);
-      }
-      int64_t far = -1;
-      while (!q.empty()) {
-        int64_t x = q.front();
-        q.pop_front();
-        far = x;
-        for (int64_t e = ip[x]; e < ip[x + 1]; ++e) {
-          int64_t y = xp[e];
-          if (dist[y] < 0) {
-            dist[y] = dist[x] + 1;
-            q.push_back(y);
-          }
-        }
-      }
-      // prefer an unreached node (disconnected component), else farthest
-      int64_t pick = -1;
-      for (int64_t x = 0; x < N; ++x) {
-        if (dist[x] < 0) {
-          pick = x;
-          break;
-        }
-      }
-      if (pick < 0) pick = far;
-      seeds.push_back(pick);
-    }
-  }
-
-  // --- balanced multi-source BFS growth: round-robin over partitions, each
-  // expands its frontier one node per turn until capacity.
-  std::vector<int64_t> psize(nparts, 0);
-  std::vector<std::deque<int64_t>> frontier(nparts);
-  for (int64_t k = 0; k < nparts; ++k) {
-    if (pp[seeds[k]] < 0) {
-      pp[seeds[k]] = static_cast<int32_t>(k);
-      psize[k]++;
-      frontier[k].push_back(seeds[k]);
-    }
-  }
-  bool progress = true;
-  while (progress) {
-    progress = false;
-    for (int64_t k = 0; k < nparts; ++k) {
-      if (psize[k] >= cap) continue;
-      // claim one unassigned neighbor from this partition's frontier
-      while (!frontier[k].empty() && psize[k] < cap) {
-        int64_t x = frontier[k].front();
-        bool claimed = false;
-        int64_t e = ip[x];
-        for (; e < ip[x + 1]; ++e) {
-          int64_t y = xp[e];
-          if (pp[y] < 0) {
-            pp[y] = static_cast<int32_t>(k);
-            psize[k]++;
-            frontier[k].push_back(y);
-            claimed = true;
-            progress = true;
-            break;
-          }
-        }
-        if (!claimed) {
-          frontier[k].pop_front();
-        } else {
-          break;  // one claim per turn keeps growth balanced
-        }
-      }
-    }
-  }
-  // leftover (disconnected / capacity-starved) nodes -> smallest partition
-  for (int64_t x = 0; x < N; ++x) {
-    if (pp[x] < 0) {
-      int64_t k =
-          std::min_element(psize.begin(), psize.end()) - psize.begin();
-      pp[x] = static_cast<int32_t>(k);
-      psize[k]++;
-    }
-  }
-
-  // --- boundary refinement (FM-lite): move boundary nodes to the
-  // neighboring partition with the best gain while keeping balance.
-  const int64_t lo =
-      static_cast<int64_t>((double)N / nparts * (1.0 - balance_slack));
-  std::vector<int32_t> cnt(nparts, 0);
-  std::vector<int32_t> touched;
-  touched.reserve(64);
-  for (int64_t pass = 0; pass < n_refine_passes; ++pass) {
-    int64_t moves = 0;
-    for (int64_t x = 0; x < N; ++x) {
-      const int32_t a = pp[x];
-      if (psize[a] <= lo) continue;
-      // count neighbors per partition
-      bool boundary = false;
-      for (int64_t e = ip[x]; e < ip[x + 1]; ++e) {
-        const int32_t q = pp[xp[e]];
-        if (cnt[q] == 0) touched.push_back(q);
-        cnt[q]++;
-        if (q != a) boundary = true;
-      }
-      if (boundary) {
-        // candidate: partition with most of x's neighbors
-        int32_t best = a;
-        int64_t best_gain = 0;
-        for (int32_t q : touched) {
-          if (q == a || psize[q] >= cap) continue;
-          int64_t gain = (int64_t)cnt[q] - (int64_t)cnt[a];
-          if (objective == 1) {
-            // volume-aware: account for x's own replica-count change
-            // (x is replicated into every foreign partition it has a
-            // neighbor in).
-            int64_t rep_a = 0, rep_q = 0;
-            for (int32_t t : touched) {
-              if (t != a && cnt[t] > 0) rep_a++;
-              if (t != q && cnt[t] > 0) rep_q++;
-            }
-            gain += (rep_a - rep_q);
-          }
-          if (gain > best_gain) {
-            best_gain = gain;
-            best = q;
-          }
-        }
-        if (best != a) {
-          pp[x] = best;
-          psize[a]--;
-          psize[best]++;
-          moves++;
-        }
-      }
-      for (int32_t q : touched) cnt[q] = 0;
-      touched.clear();
-    }
-    if (moves == 0) break;
-  }
-  return part;
-}
+// The multilevel k-way partitioner lives in partitioner.cpp.

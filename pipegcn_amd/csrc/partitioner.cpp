@@ -1,0 +1,346 @@
+// Multilevel k-way graph partitioner (METIS-style; replaces the METIS
+// library the reference used through dgl.distributed.partition_graph,
+// /root/reference/helper/utils.py:143 — METIS is not available in this
+// environment, so this is our own implementation):
+//
+//   1. COARSEN: heavy-edge matching collapses node pairs level by level
+//      (edge weights accumulate; node weights = contained fine nodes).
+//   2. INITIAL PARTITION on the coarsest graph: farthest-point seeded,
+//      weight-balanced multi-source BFS growth.
+//   3. UNCOARSEN: project the assignment back up, boundary-refining at
+//      every level (FM-style single-node moves, edge-weight gains, node
+//      -weight balance; the 'vol' objective adds the node's replica-count
+//      delta — the communication-volume term of the reference's
+//      objtype='vol').
+//
+// Exported as partition_graph_cpu with the original signature.
+
+#include "common.h"
+
+#include <algorithm>
+#include <cstring>
+#include <deque>
+#include <numeric>
+#include <random>
+#include <vector>
+
+namespace {
+
+struct Graph {
+  int64_t n = 0;
+  std::vector<int64_t> indptr;
+  std::vector<int32_t> indices;
+  std::vector<int32_t> ew;  // edge weights
+  std::vector<int64_t> nw;  // node weights (fine nodes contained)
+};
+
+// heavy-edge matching; returns coarse count and fine->coarse map
+int64_t heavy_edge_matching(const Graph& g, std::mt19937_64& rng,
+                            std::vector<int32_t>& cmap) {
+  const int64_t n = g.n;
+  std::vector<int32_t> match(n, -1);
+  std::vector<int32_t> order(n);
+  std::iota(order.begin(), order.end(), 0);
+  std::shuffle(order.begin(), order.end(), rng);
+  for (int32_t u : order) {
+    if (match[u] >= 0) continue;
+    int32_t best = -1;
+    int64_t best_w = -1;
+    for (int64_t e = g.indptr[u]; e < g.indptr[u + 1]; ++e) {
+      const int32_t v = g.indices[e];
+      if (v == u || match[v] >= 0) continue;
+      if (g.ew[e] > best_w) {
+        best_w = g.ew[e];
+        best = v;
+      }
+    }
+    if (best >= 0) {
+      match[u] = best;
+      match[best] = u;
+    } else {
+      match[u] = u;
+    }
+  }
+  cmap.assign(n, -1);
+  int64_t nc = 0;
+  for (int64_t u = 0; u < n; ++u) {
+    if (cmap[u] >= 0) continue;
+    cmap[u] = static_cast<int32_t>(nc);
+    const int32_t v = match[u];
+    if (v >= 0 && v != u && cmap[v] < 0) cmap[v] = static_cast<int32_t>(nc);
+    nc++;
+  }
+  return nc;
+}
+
+Graph coarsen(const Graph& g, const std::vector<int32_t>& cmap, int64_t nc) {
+  Graph c;
+  c.n = nc;
+  c.nw.assign(nc, 0);
+  // members of each coarse node (at most 2)
+  std::vector<int32_t> m1(nc, -1), m2(nc, -1);
+  for (int64_t u = 0; u < g.n; ++u) {
+    const int32_t cu = cmap[u];
+    c.nw[cu] += g.nw[u];
+    if (m1[cu] < 0)
+      m1[cu] = static_cast<int32_t>(u);
+    else
+      m2[cu] = static_cast<int32_t>(u);
+  }
+  c.indptr.assign(nc + 1, 0);
+  std::vector<int32_t> mark(nc, -1);
+  std::vector<int64_t> slot(nc, 0);
+  // two passes: count, then fill
+  std::vector<int32_t> tmp_idx;
+  std::vector<int32_t> tmp_w;
+  for (int64_t cu = 0; cu < nc; ++cu) {
+    tmp_idx.clear();
+    tmp_w.clear();
+    for (int32_t u : {m1[cu], m2[cu]}) {
+      if (u < 0) continue;
+      for (int64_t e = g.indptr[u]; e < g.indptr[u + 1]; ++e) {
+        const int32_t cv = cmap[g.indices[e]];
+        if (cv == cu) continue;
+        if (mark[cv] != static_cast<int32_t>(cu)) {
+          mark[cv] = static_cast<int32_t>(cu);
+          slot[cv] = tmp_idx.size();
+          tmp_idx.push_back(cv);
+          tmp_w.push_back(g.ew[e]);
+        } else {
+          tmp_w[slot[cv]] += g.ew[e];
+        }
+      }
+    }
+    c.indptr[cu + 1] = c.indptr[cu] + static_cast<int64_t>(tmp_idx.size());
+    c.indices.insert(c.indices.end(), tmp_idx.begin(), tmp_idx.end());
+    c.ew.insert(c.ew.end(), tmp_w.begin(), tmp_w.end());
+  }
+  return c;
+}
+
+// weight-balanced initial partition: farthest-point seeds + BFS growth
+void initial_partition(const Graph& g, int64_t nparts, int64_t cap_w,
+                       std::mt19937_64& rng, std::vector<int32_t>& part) {
+  const int64_t n = g.n;
+  part.assign(n, -1);
+  std::vector<int64_t> seeds;
+  {
+    std::uniform_int_distribution<int64_t> uni(0, n - 1);
+    seeds.push_back(uni(rng));
+    std::vector<int32_t> dist(n);
+    for (int64_t k = 1; k < nparts; ++k) {
+      std::fill(dist.begin(), dist.end(), -1);
+      std::deque<int64_t> q;
+      for (int64_t s : seeds) {
+        dist[s] = 0;
+        q.push_back(s);
+      }
+      int64_t far = seeds[0];
+      while (!q.empty()) {
+        const int64_t x = q.front();
+        q.pop_front();
+        far = x;
+        for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+          const int32_t y = g.indices[e];
+          if (dist[y] < 0) {
+            dist[y] = dist[x] + 1;
+            q.push_back(y);
+          }
+        }
+      }
+      int64_t pick = -1;
+      for (int64_t x = 0; x < n; ++x)
+        if (dist[x] < 0) {
+          pick = x;
+          break;
+        }
+      seeds.push_back(pick < 0 ? far : pick);
+    }
+  }
+  std::vector<int64_t> psize(nparts, 0);
+  std::vector<std::deque<int64_t>> frontier(nparts);
+  for (int64_t k = 0; k < nparts; ++k) {
+    const int64_t s = seeds[k];
+    if (part[s] < 0) {
+      part[s] = static_cast<int32_t>(k);
+      psize[k] += g.nw[s];
+      frontier[k].push_back(s);
+    }
+  }
+  bool progress = true;
+  while (progress) {
+    progress = false;
+    for (int64_t k = 0; k < nparts; ++k) {
+      if (psize[k] >= cap_w) continue;
+      while (!frontier[k].empty() && psize[k] < cap_w) {
+        const int64_t x = frontier[k].front();
+        bool claimed = false;
+        for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+          const int32_t y = g.indices[e];
+          if (part[y] < 0) {
+            part[y] = static_cast<int32_t>(k);
+            psize[k] += g.nw[y];
+            frontier[k].push_back(y);
+            claimed = true;
+            progress = true;
+            break;
+          }
+        }
+        if (!claimed)
+          frontier[k].pop_front();
+        else
+          break;  // one claim per turn keeps growth balanced
+      }
+    }
+  }
+  for (int64_t x = 0; x < n; ++x) {
+    if (part[x] < 0) {
+      const int64_t k =
+          std::min_element(psize.begin(), psize.end()) - psize.begin();
+      part[x] = static_cast<int32_t>(k);
+      psize[k] += g.nw[x];
+    }
+  }
+}
+
+// FM-style boundary refinement (single-node moves, edge-weight gains)
+void refine(const Graph& g, int64_t nparts, int64_t objective, int64_t lo_w,
+            int64_t cap_w, int64_t passes, std::vector<int32_t>& part) {
+  const int64_t n = g.n;
+  std::vector<int64_t> psize(nparts, 0);
+  for (int64_t x = 0; x < n; ++x) psize[part[x]] += g.nw[x];
+  std::vector<int64_t> cnt(nparts, 0);
+  std::vector<int32_t> touched;
+  touched.reserve(64);
+  for (int64_t pass = 0; pass < passes; ++pass) {
+    int64_t moves = 0;
+    for (int64_t x = 0; x < n; ++x) {
+      const int32_t a = part[x];
+      if (psize[a] - g.nw[x] < lo_w) continue;
+      bool boundary = false;
+      for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+        const int32_t q = part[g.indices[e]];
+        if (cnt[q] == 0) touched.push_back(q);
+        cnt[q] += g.ew[e];
+        if (q != a) boundary = true;
+      }
+      if (boundary) {
+        int32_t best = a;
+        int64_t best_gain = 0;
+        for (const int32_t q : touched) {
+          if (q == a || psize[q] + g.nw[x] > cap_w) continue;
+          int64_t gain = cnt[q] - cnt[a];
+          if (objective == 1) {
+            // communication-volume term: replica-count delta of x
+            int64_t rep_a = 0, rep_q = 0;
+            for (const int32_t t : touched) {
+              if (t != a && cnt[t] > 0) rep_a++;
+              if (t != q && cnt[t] > 0) rep_q++;
+            }
+            gain += rep_a - rep_q;
+          }
+          if (gain > best_gain) {
+            best_gain = gain;
+            best = q;
+          }
+        }
+        if (best != a) {
+          part[x] = best;
+          psize[a] -= g.nw[x];
+          psize[best] += g.nw[x];
+          moves++;
+        }
+      }
+      for (const int32_t q : touched) cnt[q] = 0;
+      touched.clear();
+    }
+    if (moves == 0) break;
+  }
+}
+
+}  // namespace
+
+torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
+                                  int64_t nparts, int64_t objective,
+                                  double balance_slack, int64_t n_refine_passes,
+                                  int64_t seed) {
+  TORCH_CHECK(nparts >= 1, "nparts must be >= 1");
+  const int64_t N = indptr.numel() - 1;
+  auto out = torch::zeros({N}, torch::kInt);
+  if (nparts == 1 || N == 0) return out;
+
+  std::mt19937_64 rng(seed);
+
+  // level 0 = the input graph, unit weights
+  std::vector<Graph> levels(1);
+  {
+    Graph& g0 = levels[0];
+    g0.n = N;
+    const int64_t* ip = indptr.data_ptr<int64_t>();
+    const int32_t* xp = indices.data_ptr<int32_t>();
+    g0.indptr.assign(ip, ip + N + 1);
+    g0.indices.assign(xp, xp + g0.indptr[N]);
+    g0.ew.assign(g0.indices.size(), 1);
+    g0.nw.assign(N, 1);
+  }
+  std::vector<std::vector<int32_t>> cmaps;
+
+  // --- coarsen
+  const int64_t coarse_target = std::max<int64_t>(128 * nparts, 4096);
+  while (levels.back().n > coarse_target && levels.size() < 24) {
+    std::vector<int32_t> cmap;
+    const int64_t nc = heavy_edge_matching(levels.back(), rng, cmap);
+    if (nc > levels.back().n * 95 / 100) break;  // matching stalled
+    levels.push_back(coarsen(levels.back(), cmap, nc));
+    cmaps.push_back(std::move(cmap));
+  }
+
+  // --- initial partition on the coarsest level
+  const int64_t tot_w = N;
+  const int64_t cap_w =
+      static_cast<int64_t>((double)tot_w / nparts * (1.0 + balance_slack)) + 1;
+  const int64_t lo_w =
+      static_cast<int64_t>((double)tot_w / nparts * (1.0 - balance_slack));
+  // Random restarts at the coarsest level (it is tiny — a few thousand
+  // nodes): greedy single-node refinement cannot escape a bad seeded
+  // growth, so try several and keep the lowest weighted edge cut.
+  std::vector<int32_t> part;
+  {
+    const Graph& gc = levels.back();
+    auto cut_of = [&](const std::vector<int32_t>& p) {
+      int64_t cut = 0;
+      for (int64_t x = 0; x < gc.n; ++x)
+        for (int64_t e = gc.indptr[x]; e < gc.indptr[x + 1]; ++e)
+          if (p[gc.indices[e]] != p[x]) cut += gc.ew[e];
+      return cut;
+    };
+    int64_t best_cut = -1;
+    for (int trial = 0; trial < 8; ++trial) {
+      std::vector<int32_t> cand;
+      initial_partition(gc, nparts, cap_w, rng, cand);
+      refine(gc, nparts, objective, lo_w, cap_w,
+             std::max<int64_t>(n_refine_passes, 64), cand);
+      const int64_t c = cut_of(cand);
+      if (best_cut < 0 || c < best_cut) {
+        best_cut = c;
+        part = std::move(cand);
+      }
+    }
+  }
+
+  // --- uncoarsen + refine each level
+  for (int64_t lvl = static_cast<int64_t>(levels.size()) - 2; lvl >= 0;
+       --lvl) {
+    const std::vector<int32_t>& cmap = cmaps[lvl];
+    std::vector<int32_t> fine(levels[lvl].n);
+    for (int64_t u = 0; u < levels[lvl].n; ++u) fine[u] = part[cmap[u]];
+    part = std::move(fine);
+    const int64_t passes =
+        lvl == 0 ? std::max<int64_t>(n_refine_passes / 2, 2)
+                 : std::max<int64_t>(n_refine_passes, 32);
+    refine(levels[lvl], nparts, objective, lo_w, cap_w, passes, part);
+  }
+
+  std::memcpy(out.data_ptr<int32_t>(), part.data(), N * sizeof(int32_t));
+  return out;
+}

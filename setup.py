@@ -20,6 +20,7 @@ ext = CUDAExtension(
     sources=[
         "pipegcn_amd/csrc/bindings.cpp",
         "pipegcn_amd/csrc/graph_core.cpp",
+        "pipegcn_amd/csrc/partitioner.cpp",
         "pipegcn_amd/csrc/hip/kernels.hip",
         "pipegcn_amd/csrc/hip/dual_gemm.hip",
     ],
