@@ -71,6 +71,12 @@ def create_parser(argv=None):
     parser.add_argument("--parts-per-node", "--parts_per_node", type=int,
                         default=10)
 
+    parser.add_argument("--dist-timeout", "--dist_timeout", type=int,
+                        default=1800,
+                        help="torch.distributed operation timeout in "
+                             "seconds — a died rank fails the job instead "
+                             "of hanging the ring forever (the reference "
+                             "hangs: SURVEY §5 failure detection)")
     parser.add_argument("--checkpoint-every", "--checkpoint_every",
                         type=int, default=0,
                         help="save a per-rank training checkpoint every N "

@@ -357,7 +357,11 @@ def init_processes(rank, size, args, device: Optional[str] = None):
     """Entry for spawned per-partition processes (reference train.py:408)."""
     os.environ["MASTER_ADDR"] = args.master_addr
     os.environ["MASTER_PORT"] = "%d" % args.port
-    dist.init_process_group(args.backend, rank=rank, world_size=size)
+    from datetime import timedelta
+
+    dist.init_process_group(
+        args.backend, rank=rank, world_size=size,
+        timeout=timedelta(seconds=getattr(args, "dist_timeout", 1800)))
     check_args(args)
     if device is None:
         device = "cuda:0" if (args.backend == "nccl"
