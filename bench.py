@@ -138,6 +138,8 @@ def main():
 
     for _ in range(args.warmup):
         step()
+    if device.startswith("cuda"):
+        torch.cuda.reset_peak_memory_stats()
     ctx.buffer.pop_comm_stats()  # reset overlap stats
     wait_s = 0.0
     barrier_sync()
@@ -156,11 +158,12 @@ def main():
     elapsed = time.time() - t0
     comm_busy = ctx.buffer.pop_comm_stats()
 
-    # whole-job epoch time = max over ranks
-    t = torch.tensor([elapsed, comm_busy, wait_s], dtype=torch.float64)
+    # whole-job epoch time = max over ranks (NCCL needs device tensors)
+    t = torch.tensor([elapsed, comm_busy, wait_s], dtype=torch.float64,
+                     device=device if backend == "nccl" else "cpu")
     if world > 1:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
-    elapsed, comm_busy, wait_s = t.tolist()
+    elapsed, comm_busy, wait_s = t.cpu().tolist()
     epoch_s = elapsed / args.steps
     overlap_pct = (100.0 * max(comm_busy - wait_s, 0.0) / comm_busy
                    if comm_busy > 1e-9 else None)
@@ -197,6 +200,9 @@ def main():
                 "comm_busy_s_per_epoch": comm_busy / args.steps,
                 "comm_wait_s_per_epoch": wait_s / args.steps,
                 "boundary_comm_overlap_pct": overlap_pct,
+                "peak_mem_gb": (round(torch.cuda.max_memory_allocated()
+                                      / 2**30, 2)
+                                if device.startswith("cuda") else None),
             },
         }
         print(json.dumps(result))
