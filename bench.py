@@ -172,8 +172,9 @@ def main():
         n, avg_deg, n_feat, n_class, _ = SHAPES[args.shape]
         baseline = 0.2660  # BASELINE.md Reddit epoch time (rank 0), other hw
         result = {
-            "metric": "epoch time (s), Reddit 3-layer GraphSAGE "
-                      "(full-graph, pipelined)",
+            "metric": (f"epoch time (s), {args.shape} "
+                       f"{args.n_layers}-layer GraphSAGE (full-graph, "
+                       f"{'pipelined' if pipeline else 'vanilla'})"),
             "value": epoch_s,
             "unit": "s/epoch",
             "n_gpus": world,
