@@ -61,3 +61,49 @@ def test_checkpoint_resume(tmp_path):
     out = run_main(tmp_path, ["--no-eval", "--checkpoint-every", "4",
                               "--resume"])
     assert "resumed from epoch 7" in out
+
+
+def test_main_yelp_style_config(tmp_path):
+    """The Yelp canonical shape: multilabel BCE, conv+linear tail layers,
+    sync batch norm, inductive, pipelined (BASELINE config #4 pattern)."""
+    out = run_main(tmp_path, ["--dataset", "synth-tinyml", "--n-layers", "3",
+                              "--n-linear", "1", "--norm", "batch",
+                              "--inductive", "--enable-pipeline"])
+    assert "Accuracy" in out  # micro-F1 eval path
+
+
+def test_multi_node_launcher(tmp_path):
+    """Simulate 2 nodes on localhost: each runs main.py with its own
+    --node-rank and --parts-per-node 1; rendezvous over MASTER_ADDR/PORT
+    (reference scripts/reddit_multi_node.sh pattern). Node 0 partitions
+    first; node 1 waits for the partition dir (shared FS assumption, same
+    as the reference)."""
+    import time
+
+    port = free_port()
+    base = [sys.executable, os.path.join(REPO, "main.py"),
+            "--dataset", "synth-tiny", "--n-partitions", "2",
+            "--parts-per-node", "1", "--n-epochs", "10", "--n-layers", "2",
+            "--n-hidden", "8", "--fix-seed", "--seed", "3", "--no-eval",
+            "--backend", "gloo", "--port", str(port),
+            "--master-addr", "127.0.0.1"]
+    env = dict(os.environ, PYTHONPATH=REPO)
+    p0 = subprocess.Popen(base + ["--node-rank", "0"], cwd=tmp_path, env=env,
+                          stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+                          text=True)
+    # node 1 skips partitioning (node_rank != 0) and needs the files;
+    # wait for node 0 to write them
+    deadline = time.time() + 60
+    meta = tmp_path / "partitions" / "synth-tiny-2-metis-vol-trans" / \
+        "meta.json"
+    while not meta.exists() and time.time() < deadline:
+        time.sleep(0.5)
+    assert meta.exists(), "node 0 never wrote the partition"
+    p1 = subprocess.Popen(base + ["--node-rank", "1"], cwd=tmp_path, env=env,
+                          stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+                          text=True)
+    out0, _ = p0.communicate(timeout=180)
+    out1, _ = p1.communicate(timeout=180)
+    assert p0.returncode == 0, out0
+    assert p1.returncode == 0, out1
+    assert "Process 000" in out0 and "Process 001" in out1
