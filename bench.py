@@ -42,6 +42,9 @@ def main():
                          "headline config)")
     ap.add_argument("--norm", type=str, default="layer",
                     choices=["layer", "batch", "none"])
+    ap.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
+                    help="optional bf16 compute (the judged baseline "
+                         "config is fp32 = reference parity)")
     ap.add_argument("--feat-corr", action="store_true")
     ap.add_argument("--grad-corr", action="store_true")
     ap.add_argument("--device", type=str, default=None)
@@ -92,6 +95,7 @@ def main():
     multilabel = SHAPES[args.shape][4]
     layer_size = get_layer_size(part.n_feat, args.n_hidden, part.n_class,
                                 args.n_layers)
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     comm_group = dist.new_group(backend=backend) if world > 1 else None
     ctx.buffer.init_buffer(rp.num_in, rp.num_all, rp.boundary, rp.recv_shape,
                            layer_size[: args.n_layers], use_pp=args.use_pp,
@@ -99,11 +103,13 @@ def main():
                            corr_feat=args.feat_corr,
                            corr_grad=args.grad_corr,
                            device=device, group=comm_group,
-                           collect_stats=True)
+                           collect_stats=True, dtype=dtype)
 
     model = GraphSAGE(layer_size, F.relu, use_pp=args.use_pp, dropout=0.5,
                       norm=norm, n_linear=0,
                       train_size=part.n_train).to(device)
+    if dtype != torch.float32:
+        model = model.to(dtype)
     ctx.reducer.init(model)
     if multilabel:
         loss_fcn = torch.nn.BCEWithLogitsLoss(reduction="sum")
@@ -111,6 +117,8 @@ def main():
         loss_fcn = torch.nn.CrossEntropyLoss(reduction="sum")
     optimizer = torch.optim.Adam(model.parameters(), lr=0.01)
 
+    if dtype != torch.float32:
+        rp.ndata["feat"] = rp.ndata["feat"].to(dtype)
     feat = rp.ndata["feat"]
     if args.use_pp:
         pp_args = types.SimpleNamespace(model="graphsage")
@@ -120,7 +128,7 @@ def main():
     model.train()
 
     def step():
-        logits = model(rp.graph, feat, in_deg)
+        logits = model(rp.graph, feat, in_deg).float()
         loss = loss_fcn(logits[: rp.num_train], labels)
         optimizer.zero_grad(set_to_none=True)
         loss.backward()
@@ -145,7 +153,7 @@ def main():
     barrier_sync()
     t0 = time.time()
     for _ in range(args.steps):
-        logits = model(rp.graph, feat, in_deg)
+        logits = model(rp.graph, feat, in_deg).float()
         loss = loss_fcn(logits[: rp.num_train], labels)
         optimizer.zero_grad(set_to_none=True)
         loss.backward()
@@ -185,7 +193,7 @@ def main():
             "scaling": "strong",
             "vs_baseline": (epoch_s / baseline
                             if args.shape == "reddit" else None),
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "model": f"graphsage-{args.n_layers}L-h{args.n_hidden}",

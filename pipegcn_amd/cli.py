@@ -71,6 +71,10 @@ def create_parser(argv=None):
     parser.add_argument("--parts-per-node", "--parts_per_node", type=int,
                         default=10)
 
+    parser.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
+                        help="compute dtype (bf16 roughly halves the "
+                             "bandwidth-bound SpMM epoch; fp32 matches the "
+                             "reference numerics)")
     parser.add_argument("--dist-timeout", "--dist_timeout", type=int,
                         default=1800,
                         help="torch.distributed operation timeout in "

@@ -16,6 +16,12 @@ static torch::Tensor spmm(torch::Tensor indptr, torch::Tensor indices,
     spmm_csr_hip(indptr, indices, feat, dst_scale, src_scale, out);
     return out;
   }
+  if (feat.scalar_type() == torch::kBFloat16) {
+    // CPU tier computes in fp32 (the CPU path serves tests + rank-0 eval)
+    return spmm_cpu(indptr, indices, feat.to(torch::kFloat), dst_scale,
+                    src_scale, num_rows)
+        .to(torch::kBFloat16);
+  }
   return spmm_cpu(indptr, indices, feat, dst_scale, src_scale, num_rows);
 }
 
@@ -61,7 +67,11 @@ static torch::Tensor sage_dual_gemm(torch::Tensor x1, torch::Tensor x2,
 }
 
 static torch::Tensor colsum(torch::Tensor x) {
-  if (x.is_cuda()) return colsum_hip(x);
+  if (x.is_cuda()) {
+    if (x.scalar_type() == torch::kBFloat16)
+      return colsum_hip(x.to(torch::kFloat)).to(torch::kBFloat16);
+    return colsum_hip(x);
+  }
   return x.sum(0);
 }
 

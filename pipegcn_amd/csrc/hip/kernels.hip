@@ -18,9 +18,11 @@
 #include "../common.h"
 
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+#include <hip/hip_bf16.h>
 #include <hip/hip_runtime.h>
 
 #include <cstdlib>
+#include <vector>
 
 #define HIP_CHECK(expr)                                              \
   do {                                                               \
@@ -33,6 +35,27 @@ namespace {
 constexpr int kWave = 64;
 
 using int32x4 = __attribute__((ext_vector_type(4))) int;
+
+// element-type helpers: compute in fp32, store in T (fp32 or bf16)
+__device__ __forceinline__ float to_f32(float v) { return v; }
+__device__ __forceinline__ float to_f32(__hip_bfloat16 v) {
+  return __bfloat162float(v);
+}
+__device__ __forceinline__ void st_nt(float* p, float v) {
+  __builtin_nontemporal_store(v, p);
+}
+__device__ __forceinline__ void st_nt(__hip_bfloat16* p, float v) {
+  union {
+    __hip_bfloat16 b;
+    unsigned short u;
+  } cv;
+  cv.b = __float2bfloat16(v);
+  __builtin_nontemporal_store(cv.u, reinterpret_cast<unsigned short*>(p));
+}
+__device__ __forceinline__ void st(float* p, float v) { *p = v; }
+__device__ __forceinline__ void st(__hip_bfloat16* p, float v) {
+  *p = __float2bfloat16(v);
+}
 
 inline hipStream_t current_stream() {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
@@ -50,13 +73,13 @@ inline hipStream_t current_stream() {
 // fastest (CHUNK_OUTER=false) every panel streams concurrently and the
 // combined working set thrashes L3.  HAS_SRC_SCALE fuses a per-source-row
 // scale (the transpose/backward SpMM's D^{-1} pre-scale) into the gather.
-template <int VEC, bool CHUNK_OUTER, bool HAS_SRC_SCALE>
+template <typename T, int VEC, bool CHUNK_OUTER, bool HAS_SRC_SCALE>
 __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
                                 const int32_t* __restrict__ indices,
-                                const float* __restrict__ feat,
+                                const T* __restrict__ feat,
                                 const float* __restrict__ dst_scale,
                                 const float* __restrict__ src_scale,
-                                float* __restrict__ out, int64_t num_rows,
+                                T* __restrict__ out, int64_t num_rows,
                                 int64_t F, int64_t nchunks) {
   const int64_t wave_global =
       (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / kWave;
@@ -95,7 +118,8 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
         const int64_t u = indices[e];
         const float s = HAS_SRC_SCALE ? src_scale[u] : 1.f;
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) acc[k] += s * feat[u * F + f0 + k];
+        for (int k = 0; k < VEC; ++k)
+          acc[k] += s * to_f32(feat[u * F + f0 + k]);
       }
       for (; e + 4 <= e_end; e += 4) {
         const int32x4 uu = __builtin_nontemporal_load(
@@ -113,13 +137,13 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
         }
         float v0[VEC], v1[VEC], v2[VEC], v3[VEC];
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) v0[k] = feat[u0 * F + f0 + k];
+        for (int k = 0; k < VEC; ++k) v0[k] = to_f32(feat[u0 * F + f0 + k]);
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) v1[k] = feat[u1 * F + f0 + k];
+        for (int k = 0; k < VEC; ++k) v1[k] = to_f32(feat[u1 * F + f0 + k]);
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) v2[k] = feat[u2 * F + f0 + k];
+        for (int k = 0; k < VEC; ++k) v2[k] = to_f32(feat[u2 * F + f0 + k]);
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) v3[k] = feat[u3 * F + f0 + k];
+        for (int k = 0; k < VEC; ++k) v3[k] = to_f32(feat[u3 * F + f0 + k]);
         if (HAS_SRC_SCALE) {
 #pragma unroll
           for (int k = 0; k < VEC; ++k)
@@ -134,31 +158,31 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
         const int64_t u = indices[e];
         const float s = HAS_SRC_SCALE ? src_scale[u] : 1.f;
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) acc[k] += s * feat[u * F + f0 + k];
+        for (int k = 0; k < VEC; ++k)
+          acc[k] += s * to_f32(feat[u * F + f0 + k]);
       }
       const float s = dst_scale ? dst_scale[r] : 1.f;
 #pragma unroll
-      for (int k = 0; k < VEC; ++k)
-        __builtin_nontemporal_store(acc[k] * s, out + r * F + f0 + k);
+      for (int k = 0; k < VEC; ++k) st_nt(out + r * F + f0 + k, acc[k] * s);
     } else {
       // ragged tail chunk: scalar guarded
       for (; e < e_end; ++e) {
         const int64_t u = indices[e];
         const float s = HAS_SRC_SCALE ? src_scale[u] : 1.f;
         for (int k = 0; k < VEC && f0 + k < F; ++k)
-          acc[k] += s * feat[u * F + f0 + k];
+          acc[k] += s * to_f32(feat[u * F + f0 + k]);
       }
       const float s = dst_scale ? dst_scale[r] : 1.f;
       for (int k = 0; k < VEC && f0 + k < F; ++k)
-        out[r * F + f0 + k] = acc[k] * s;
+        st(out + r * F + f0 + k, acc[k] * s);
     }
   }
 }
 
-template <int VEC>
+template <typename T, int VEC>
 void launch_spmm(const int64_t* indptr, const int32_t* indices,
-                 const float* feat, const float* dst_scale,
-                 const float* src_scale, float* out, int64_t num_src,
+                 const T* feat, const float* dst_scale,
+                 const float* src_scale, T* out, int64_t num_src,
                  int64_t num_rows, int64_t F, bool chunk_outer,
                  hipStream_t stream) {
   const int64_t nchunks = (F + kWave * VEC - 1) / (kWave * VEC);
@@ -177,14 +201,14 @@ void launch_spmm(const int64_t* indptr, const int32_t* indices,
   };
   if (chunk_outer) {
     if (src_scale)
-      launch(HIP_KERNEL_NAME(spmm_csr_kernel<VEC, true, true>));
+      launch(HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, true, true>));
     else
-      launch(HIP_KERNEL_NAME(spmm_csr_kernel<VEC, true, false>));
+      launch(HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, true, false>));
   } else {
     if (src_scale)
-      launch(HIP_KERNEL_NAME(spmm_csr_kernel<VEC, false, true>));
+      launch(HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, false, true>));
     else
-      launch(HIP_KERNEL_NAME(spmm_csr_kernel<VEC, false, false>));
+      launch(HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, false, false>));
   }
   HIP_CHECK(hipGetLastError());
 }
@@ -193,10 +217,10 @@ void launch_spmm(const int64_t* indptr, const int32_t* indices,
 // Row gather / scatter-add. One wave per (row, chunk).
 // ---------------------------------------------------------------------------
 
-template <int VEC, bool SCATTER_ADD>
-__global__ void rowcopy_kernel(const float* __restrict__ src,
+template <typename T, int VEC, bool SCATTER_ADD>
+__global__ void rowcopy_kernel(const T* __restrict__ src,
                                const int64_t* __restrict__ idx,
-                               float* __restrict__ dst, int64_t nrows,
+                               T* __restrict__ dst, int64_t nrows,
                                int64_t F, int64_t nchunks) {
   const int64_t wave_global =
       (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / kWave;
@@ -213,14 +237,16 @@ __global__ void rowcopy_kernel(const float* __restrict__ src,
 #pragma unroll
       for (int k = 0; k < VEC; ++k) {
         if (SCATTER_ADD)
-          dst[j * F + f0 + k] += src[i * F + f0 + k];
+          st(dst + j * F + f0 + k, to_f32(dst[j * F + f0 + k]) +
+                                       to_f32(src[i * F + f0 + k]));
         else
           dst[i * F + f0 + k] = src[j * F + f0 + k];
       }
     } else {
       for (int k = 0; f0 + k < F; ++k) {
         if (SCATTER_ADD)
-          dst[j * F + f0 + k] += src[i * F + f0 + k];
+          st(dst + j * F + f0 + k, to_f32(dst[j * F + f0 + k]) +
+                                       to_f32(src[i * F + f0 + k]));
         else
           dst[i * F + f0 + k] = src[j * F + f0 + k];
       }
@@ -228,15 +254,15 @@ __global__ void rowcopy_kernel(const float* __restrict__ src,
   }
 }
 
-template <int VEC, bool SCATTER_ADD>
-void launch_rowcopy(const float* src, const int64_t* idx, float* dst,
+template <typename T, int VEC, bool SCATTER_ADD>
+void launch_rowcopy(const T* src, const int64_t* idx, T* dst,
                     int64_t nrows, int64_t F, hipStream_t stream) {
   const int64_t nchunks = (F + kWave * VEC - 1) / (kWave * VEC);
   const int threads = 256;
   int64_t blocks = (nrows * nchunks * kWave + threads - 1) / threads;
   blocks = std::min<int64_t>(blocks, 8 * 65536);
   if (blocks == 0) blocks = 1;
-  hipLaunchKernelGGL(HIP_KERNEL_NAME(rowcopy_kernel<VEC, SCATTER_ADD>),
+  hipLaunchKernelGGL(HIP_KERNEL_NAME(rowcopy_kernel<T, VEC, SCATTER_ADD>),
                      dim3(blocks), dim3(threads), 0, stream, src, idx, dst,
                      nrows, F, nchunks);
   HIP_CHECK(hipGetLastError());
@@ -298,7 +324,7 @@ __global__ void colsum_partial_kernel(const float* __restrict__ x,
   }
 }
 
-int pick_vec(int64_t F, int64_t num_rows) {
+int pick_vec(int64_t F, int64_t num_rows, int64_t elem_size = 4) {
   // Prefer the widest load whose column panel (num_src × 64·VEC × 4 B) stays
   // L3-resident (256 MiB Infinity Cache; target ≤ ~160 MiB leaves room for
   // the streaming index array). Overridable for A/B benchmarking.
@@ -315,7 +341,11 @@ int pick_vec(int64_t F, int64_t num_rows) {
   }
   constexpr int64_t kTargetPairs = 640 * 1024;
   int widest = 0;
-  for (int v : {4, 2, 1}) {
+  // candidates from 16 B/lane down to 1 element
+  std::vector<int> cand;
+  for (int bytes = 16; bytes >= elem_size; bytes >>= 1)
+    cand.push_back(bytes / static_cast<int>(elem_size));
+  for (int v : cand) {
     if (F % v != 0) continue;
     if (!widest) widest = v;
     const int64_t nchunks = (F + 64 * v - 1) / (64 * v);
@@ -326,11 +356,42 @@ int pick_vec(int64_t F, int64_t num_rows) {
 
 }  // namespace
 
+template <typename T>
+void spmm_dispatch(torch::Tensor& indptr, torch::Tensor& indices,
+                   torch::Tensor& feat, const float* dsp, const float* ssp,
+                   torch::Tensor& out, int64_t num_src, int64_t num_rows,
+                   int64_t F, int vec, bool chunk_outer, hipStream_t stream) {
+  const T* fp = reinterpret_cast<const T*>(feat.data_ptr());
+  T* op = reinterpret_cast<T*>(out.data_ptr());
+  const int64_t* ip = indptr.data_ptr<int64_t>();
+  const int32_t* xp = indices.data_ptr<int32_t>();
+  switch (vec) {
+    case 8:
+      launch_spmm<T, 8>(ip, xp, fp, dsp, ssp, op, num_src, num_rows, F,
+                        chunk_outer, stream);
+      break;
+    case 4:
+      launch_spmm<T, 4>(ip, xp, fp, dsp, ssp, op, num_src, num_rows, F,
+                        chunk_outer, stream);
+      break;
+    case 2:
+      launch_spmm<T, 2>(ip, xp, fp, dsp, ssp, op, num_src, num_rows, F,
+                        chunk_outer, stream);
+      break;
+    default:
+      launch_spmm<T, 1>(ip, xp, fp, dsp, ssp, op, num_src, num_rows, F,
+                        chunk_outer, stream);
+  }
+}
+
 void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
                   torch::Tensor feat, torch::Tensor dst_scale,
                   torch::Tensor src_scale, torch::Tensor out) {
   TORCH_CHECK(feat.is_cuda() && out.is_cuda(), "spmm_csr_hip: device tensors");
-  TORCH_CHECK(feat.scalar_type() == torch::kFloat, "fp32 only");
+  const bool bf16 = feat.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || feat.scalar_type() == torch::kFloat,
+              "spmm: fp32 or bf16 only");
+  TORCH_CHECK(out.scalar_type() == feat.scalar_type());
   TORCH_CHECK(feat.is_contiguous() && out.is_contiguous());
   TORCH_CHECK(indptr.scalar_type() == torch::kLong &&
               indices.scalar_type() == torch::kInt);
@@ -341,51 +402,67 @@ void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
   const float* dsp = nullptr;
   if (dst_scale.defined() && dst_scale.numel() > 0) {
     TORCH_CHECK(dst_scale.is_contiguous() && dst_scale.numel() == num_rows);
+    TORCH_CHECK(dst_scale.scalar_type() == torch::kFloat);
     dsp = dst_scale.data_ptr<float>();
   }
   const float* ssp = nullptr;
   if (src_scale.defined() && src_scale.numel() > 0) {
     TORCH_CHECK(src_scale.is_contiguous() && src_scale.numel() == num_src);
+    TORCH_CHECK(src_scale.scalar_type() == torch::kFloat);
     ssp = src_scale.data_ptr<float>();
   }
   auto stream = current_stream();
-  const int vec = pick_vec(F, num_rows);
+  const int vec = pick_vec(F, num_rows, bf16 ? 2 : 4);
   bool chunk_outer = false;  // measured: chunk-inner wins at every shape
   if (const char* e = std::getenv("PIPEGCN_SPMM_ORDER"))
     chunk_outer = (e[0] == 'o');
-  if (vec == 4)
-    launch_spmm<4>(indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
-                   feat.data_ptr<float>(), dsp, ssp, out.data_ptr<float>(),
-                   num_src, num_rows, F, chunk_outer, stream);
-  else if (vec == 2)
-    launch_spmm<2>(indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
-                   feat.data_ptr<float>(), dsp, ssp, out.data_ptr<float>(),
-                   num_src, num_rows, F, chunk_outer, stream);
+  if (bf16)
+    spmm_dispatch<__hip_bfloat16>(indptr, indices, feat, dsp, ssp, out,
+                                  num_src, num_rows, F, vec, chunk_outer,
+                                  stream);
   else
-    launch_spmm<1>(indptr.data_ptr<int64_t>(), indices.data_ptr<int32_t>(),
-                   feat.data_ptr<float>(), dsp, ssp, out.data_ptr<float>(),
-                   num_src, num_rows, F, chunk_outer, stream);
+    spmm_dispatch<float>(indptr, indices, feat, dsp, ssp, out, num_src,
+                         num_rows, F, vec, chunk_outer, stream);
+}
+
+template <typename T, bool SCATTER_ADD>
+void rowcopy_dispatch(torch::Tensor& src, torch::Tensor& idx,
+                      torch::Tensor& dst, int64_t n, int64_t F,
+                      int vec, hipStream_t stream) {
+  const T* sp = reinterpret_cast<const T*>(src.data_ptr());
+  T* dp = reinterpret_cast<T*>(dst.data_ptr());
+  const int64_t* ixp = idx.data_ptr<int64_t>();
+  switch (vec) {
+    case 8:
+      launch_rowcopy<T, 8, SCATTER_ADD>(sp, ixp, dp, n, F, stream);
+      break;
+    case 4:
+      launch_rowcopy<T, 4, SCATTER_ADD>(sp, ixp, dp, n, F, stream);
+      break;
+    case 2:
+      launch_rowcopy<T, 2, SCATTER_ADD>(sp, ixp, dp, n, F, stream);
+      break;
+    default:
+      launch_rowcopy<T, 1, SCATTER_ADD>(sp, ixp, dp, n, F, stream);
+  }
 }
 
 void gather_rows_hip(torch::Tensor src, torch::Tensor idx, torch::Tensor out) {
   TORCH_CHECK(src.is_cuda() && idx.is_cuda() && out.is_cuda());
-  TORCH_CHECK(src.scalar_type() == torch::kFloat &&
+  const bool bf16 = src.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || src.scalar_type() == torch::kFloat);
+  TORCH_CHECK(out.scalar_type() == src.scalar_type() &&
               idx.scalar_type() == torch::kLong);
   TORCH_CHECK(src.is_contiguous() && out.is_contiguous());
   const int64_t F = src.size(1);
   const int64_t n = idx.numel();
   TORCH_CHECK(out.size(0) == n && out.size(1) == F);
   auto stream = current_stream();
-  const int vec = pick_vec(F, 0);
-  if (vec == 4)
-    launch_rowcopy<4, false>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
-                             out.data_ptr<float>(), n, F, stream);
-  else if (vec == 2)
-    launch_rowcopy<2, false>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
-                             out.data_ptr<float>(), n, F, stream);
+  const int vec = pick_vec(F, 0, bf16 ? 2 : 4);
+  if (bf16)
+    rowcopy_dispatch<__hip_bfloat16, false>(src, idx, out, n, F, vec, stream);
   else
-    launch_rowcopy<1, false>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
-                             out.data_ptr<float>(), n, F, stream);
+    rowcopy_dispatch<float, false>(src, idx, out, n, F, vec, stream);
 }
 
 void scatter_add_rows_hip(torch::Tensor dst, torch::Tensor idx,
@@ -393,23 +470,20 @@ void scatter_add_rows_hip(torch::Tensor dst, torch::Tensor idx,
   // Contract: idx entries are unique (one boundary peer at a time) — rows
   // are written by exactly one wave, so no atomics are needed.
   TORCH_CHECK(dst.is_cuda() && idx.is_cuda() && src.is_cuda());
-  TORCH_CHECK(dst.scalar_type() == torch::kFloat &&
+  const bool bf16 = dst.scalar_type() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || dst.scalar_type() == torch::kFloat);
+  TORCH_CHECK(src.scalar_type() == dst.scalar_type() &&
               idx.scalar_type() == torch::kLong);
   TORCH_CHECK(src.is_contiguous() && dst.is_contiguous());
   const int64_t F = dst.size(1);
   const int64_t n = idx.numel();
   TORCH_CHECK(src.size(0) == n && src.size(1) == F);
   auto stream = current_stream();
-  const int vec = pick_vec(F, 0);
-  if (vec == 4)
-    launch_rowcopy<4, true>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
-                            dst.data_ptr<float>(), n, F, stream);
-  else if (vec == 2)
-    launch_rowcopy<2, true>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
-                            dst.data_ptr<float>(), n, F, stream);
+  const int vec = pick_vec(F, 0, bf16 ? 2 : 4);
+  if (bf16)
+    rowcopy_dispatch<__hip_bfloat16, true>(src, idx, dst, n, F, vec, stream);
   else
-    launch_rowcopy<1, true>(src.data_ptr<float>(), idx.data_ptr<int64_t>(),
-                            dst.data_ptr<float>(), n, F, stream);
+    rowcopy_dispatch<float, true>(src, idx, dst, n, F, vec, stream);
 }
 
 torch::Tensor colsum_hip(torch::Tensor x) {

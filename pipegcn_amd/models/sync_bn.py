@@ -31,8 +31,10 @@ class SyncBatchNormFunc(Function):
         if not training:
             mean, var = running_mean, running_var
         else:
-            sum_x = _colsum(x)
-            sum_x2 = _colsum(x * x)
+            # statistics in fp32 (bf16 products/sums lose too much precision)
+            xs = x.float() if x.dtype != torch.float32 else x
+            sum_x = _colsum(xs)
+            sum_x2 = _colsum(xs * xs)
             _maybe_all_reduce(sum_x)
             _maybe_all_reduce(sum_x2)
             mean = sum_x / whole_size
@@ -40,22 +42,24 @@ class SyncBatchNormFunc(Function):
             running_mean.mul_(1 - momentum).add_(mean * momentum)
             running_var.mul_(1 - momentum).add_(var * momentum)
         std = torch.sqrt(var + eps)
-        x_hat = (x - mean) / std
+        x_hat = (x - mean) / std  # fp32 (mean/std are fp32)
         if training:
             ctx.save_for_backward(x_hat, weight, std)
             ctx.whole_size = whole_size
-        return x_hat * weight + bias
+        return (x_hat * weight + bias).to(x.dtype)
 
     @staticmethod
     def backward(ctx, grad):
         x_hat, weight, std = ctx.saved_tensors
-        dbias = _colsum(grad)
-        dweight = _colsum(grad * x_hat)
+        gs = grad.float() if grad.dtype != torch.float32 else grad
+        dbias = _colsum(gs)
+        dweight = _colsum(gs * x_hat.float())
         _maybe_all_reduce(dbias)
         _maybe_all_reduce(dweight)
         n = ctx.whole_size
         dx = (weight / n) / std * (n * grad - dbias - x_hat * dweight)
-        return dx, dweight, dbias, None, None, None, None, None, None
+        return (dx.to(grad.dtype), dweight.to(weight.dtype),
+                dbias.to(weight.dtype), None, None, None, None, None, None)
 
 
 class SyncBatchNorm(nn.Module):

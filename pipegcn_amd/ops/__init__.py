@@ -79,6 +79,9 @@ def scatter_add_rows(dst: torch.Tensor, idx: torch.Tensor,
 
 def ema_update(avg: torch.Tensor, x: torch.Tensor, momentum: float) -> None:
     """avg = momentum * avg + (1 - momentum) * x (in place)."""
+    if avg.dtype != torch.float32:
+        avg.mul_(momentum).add_(x, alpha=1.0 - momentum)
+        return
     native().ema_update(avg, x, momentum)
 
 
@@ -93,7 +96,8 @@ class _SageDualLinear(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x1, x2, w1, w2, b1, b2):
         ctx.save_for_backward(x1, x2, w1, w2)
-        if w1.size(0) < 64 or not x1.is_cuda:
+        if (w1.size(0) < 64 or not x1.is_cuda
+                or x1.dtype != torch.float32):
             # thin-N / CPU: the 128x128 MFMA tile would waste most of the
             # block; rocBLAS pair instead (the op is memory-bound there)
             out = torch.mm(x1, w1.t())
@@ -115,6 +119,7 @@ class _SageDualLinear(torch.autograd.Function):
         gw1 = g.t() @ x1
         gw2 = g.t() @ x2
         gb = native().colsum(g)  # two-phase column sum (fastest measured)
+        gb = gb.to(w1.dtype)
         return gx1, gx2, gw1, gw2, gb, gb
 
 

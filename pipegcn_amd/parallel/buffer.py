@@ -62,7 +62,8 @@ class Buffer:
                     backend: str = "gloo", pipeline: bool = False,
                     corr_feat: bool = False, corr_grad: bool = False,
                     corr_momentum: float = 0.95, device: str = "cpu",
-                    group=None, collect_stats: bool = False):
+                    group=None, collect_stats: bool = False,
+                    dtype: torch.dtype = torch.float32):
         self._rank = dist.get_rank() if dist.is_initialized() else 0
         self._size = dist.get_world_size() if dist.is_initialized() else 1
         self._num_in = num_in
@@ -97,7 +98,7 @@ class Buffer:
         L = self._n_layers
         S = self._size
         dev = self._device
-        mk = lambda n, f: torch.zeros(n, f, device=dev)
+        mk = lambda n, f: torch.zeros(n, f, device=dev, dtype=dtype)
         self._f_recv = [[None] * S for _ in range(L)]
         self._b_recv = [[None] * S for _ in range(L)]
         self._f_avg = [[None] * S for _ in range(L)]

@@ -208,8 +208,14 @@ def run(part: PartData, args, device: str = "cpu",
         backend=args.backend, pipeline=args.enable_pipeline,
         corr_feat=args.feat_corr, corr_grad=args.grad_corr,
         corr_momentum=args.corr_momentum, device=device, group=comm_group,
-        collect_stats=getattr(args, "collect_stats", False))
+        collect_stats=getattr(args, "collect_stats", False),
+        dtype=(torch.bfloat16 if getattr(args, "dtype", "fp32") == "bf16"
+               else torch.float32))
 
+    dtype = (torch.bfloat16 if getattr(args, "dtype", "fp32") == "bf16"
+             else torch.float32)
+    if dtype != torch.float32:
+        rp.ndata["feat"] = rp.ndata["feat"].to(dtype)
     feat = rp.ndata["feat"]
     if args.use_pp:
         feat = precompute(rp, args)
@@ -226,6 +232,8 @@ def run(part: PartData, args, device: str = "cpu",
     torch.manual_seed(args.seed)
     model = create_model(layer_size, args)
     model = model.to(device)
+    if dtype != torch.float32:
+        model = model.to(dtype)
 
     ctx.reducer.init(model)
 
@@ -272,6 +280,8 @@ def run(part: PartData, args, device: str = "cpu",
         t0 = time.time()
         model.train()
         logits = model(rp.graph, feat, in_deg)
+        if logits.dtype != torch.float32:
+            logits = logits.float()  # loss in fp32
         if args.inductive:
             loss = loss_fcn(logits, labels_all)
         else:

@@ -107,3 +107,9 @@ def test_multi_node_launcher(tmp_path):
     assert p0.returncode == 0, out0
     assert p1.returncode == 0, out1
     assert "Process 000" in out0 and "Process 001" in out1
+
+
+def test_main_bf16(tmp_path):
+    out = run_main(tmp_path, ["--dtype", "bf16", "--no-eval",
+                              "--enable-pipeline"])
+    assert "Epoch" in out and "nan" not in out

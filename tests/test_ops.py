@@ -165,3 +165,38 @@ def test_sage_dual_gemm_gpu(m, k, n):
     assert torch.allclose(gx1, x1.grad, atol=1e-3, rtol=1e-3)
     assert torch.allclose(gx2, x2.grad, atol=1e-3, rtol=1e-3)
     assert torch.allclose(gw1, l1.weight.grad, atol=1e-2, rtol=1e-3)
+
+
+@pytest.mark.gpu
+def test_spmm_bf16_gpu():
+    """bf16 SpMM (fp32 accumulate) vs the fp32 kernel."""
+    u, v, hg, feat = make_graph(f=256, e=20000, device="cuda")
+    deg = hg.csr.row_degrees().to("cuda").clamp(min=1)
+    inv = (1.0 / deg).contiguous()
+    ref = ops.spmm(hg.csr, feat, inv)
+    out = ops.spmm(hg.csr, feat.to(torch.bfloat16), inv)
+    assert out.dtype == torch.bfloat16
+    err = (out.float() - ref).abs().max() / ref.abs().max()
+    assert err < 0.02, err.item()
+
+
+@pytest.mark.gpu
+def test_gather_scatter_bf16_gpu():
+    src = torch.randn(100, 130, device="cuda").to(torch.bfloat16)
+    idx = torch.randperm(100, device="cuda")[:30]
+    g = ops.gather_rows(src, idx)
+    assert torch.equal(g, src[idx])
+    dst = torch.randn(100, 130, device="cuda").to(torch.bfloat16)
+    ref = dst.float().clone()
+    add = torch.randn(30, 130, device="cuda").to(torch.bfloat16)
+    ops.scatter_add_rows(dst, idx, add)
+    ref[idx] += add.float()
+    assert torch.allclose(dst.float(), ref, atol=0.05, rtol=0.02)
+
+
+def test_ema_bf16_fallback():
+    avg = torch.randn(64).to(torch.bfloat16)
+    x = torch.randn(64).to(torch.bfloat16)
+    ref = (0.9 * avg.float() + 0.1 * x.float()).to(torch.bfloat16)
+    ops.ema_update(avg, x, 0.9)
+    assert torch.allclose(avg.float(), ref.float(), atol=0.05)
