@@ -337,13 +337,17 @@ int pick_vec(int64_t F, int64_t num_rows, int64_t elem_size = 4) {
   // reaches ~40 per resident wave (16384 waves -> 640k pairs), else widest.
   if (const char* e = std::getenv("PIPEGCN_SPMM_VEC")) {
     int v = std::atoi(e);
-    if ((v == 4 || v == 2 || v == 1) && F % v == 0) return v;
+    if ((v == 8 || v == 4 || v == 2 || v == 1) && F % v == 0 &&
+        v * elem_size <= 16)
+      return v;
   }
   constexpr int64_t kTargetPairs = 640 * 1024;
   int widest = 0;
   // candidates from 16 B/lane down to 1 element
+  // lane loads below 4 B are never worth it (bf16 VEC1 measured 40% slower
+  // than VEC4 at F=256)
   std::vector<int> cand;
-  for (int bytes = 16; bytes >= elem_size; bytes >>= 1)
+  for (int bytes = 16; bytes >= std::max<int64_t>(elem_size, 4); bytes >>= 1)
     cand.push_back(bytes / static_cast<int>(elem_size));
   for (int v : cand) {
     if (F % v != 0) continue;

@@ -28,6 +28,7 @@ def build(n=232_965, avg_deg=492, seed=0):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--f", type=int, nargs="+", default=[602, 256])
+    ap.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32")
     ap.add_argument("--nodes", type=int, default=232_965)
     ap.add_argument("--deg", type=int, default=492)
     ap.add_argument("--rounds", type=int, default=5)
@@ -42,11 +43,14 @@ def main():
     inv = (1.0 / deg).contiguous()
     print(f"graph: {n} nodes, {E} edges")
 
+    elem = 2 if args.dtype == "bf16" else 4
     for F in args.f:
         feat = torch.randn(n, F, device="cuda")
+        if args.dtype == "bf16":
+            feat = feat.to(torch.bfloat16)
         variants = []
-        for vec in (4, 2, 1):
-            if F % vec:
+        for vec in (8, 4, 2, 1):
+            if F % vec or vec * elem > 16:
                 continue
             for order in ("o", "i"):
                 variants.append((vec, order))
@@ -60,7 +64,8 @@ def main():
                 if ref is None:
                     ref = out.clone()
                 else:
-                    assert torch.allclose(out, ref, atol=1e-4), \
+                    assert torch.allclose(out.float(), ref.float(),
+                                          atol=1e-4), \
                         f"variant {vec}{order} WRONG"
                 torch.cuda.synchronize()
                 t0 = time.time()
@@ -71,7 +76,7 @@ def main():
         print(f"F={F}:")
         for (vec, order), ts in results.items():
             ms = min(ts) * 1e3
-            gbs = E * F * 4 / min(ts) / 1e9
+            gbs = E * F * elem / min(ts) / 1e9
             tag = "chunk-outer" if order == "o" else "chunk-inner"
             print(f"  VEC={vec} {tag:12s}: {ms:8.2f} ms  "
                   f"{gbs:8.0f} GB/s logical")
