@@ -50,7 +50,8 @@ def evaluate_trans(name, model, g: FullGraph, result_file_name=None):
     model.eval()
     model.cpu()
     feat, labels = g.ndata["feat"], g.ndata["label"]
-    logits = model(g, feat)
+    feat = feat.to(next(model.parameters()).dtype)
+    logits = model(g, feat).float()
     val_acc = calc_acc(logits[g.ndata["val_mask"]],
                        labels[g.ndata["val_mask"]])
     test_acc = calc_acc(logits[g.ndata["test_mask"]],
@@ -70,8 +71,9 @@ def evaluate_induc(name, model, g: FullGraph, mode, result_file_name=None):
     model.eval()
     model.cpu()
     feat, labels = g.ndata["feat"], g.ndata["label"]
+    feat = feat.to(next(model.parameters()).dtype)
     mask = g.ndata[mode + "_mask"]
-    logits = model(g, feat)
+    logits = model(g, feat).float()
     acc = calc_acc(logits[mask], labels[mask])
     buf = "{:s} | Accuracy {:.2%}".format(name, acc)
     if result_file_name is not None:
