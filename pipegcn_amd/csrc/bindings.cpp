@@ -10,10 +10,12 @@
 
 static torch::Tensor spmm(torch::Tensor indptr, torch::Tensor indices,
                           torch::Tensor feat, torch::Tensor dst_scale,
-                          torch::Tensor src_scale, int64_t num_rows) {
+                          torch::Tensor src_scale, torch::Tensor row_order,
+                          int64_t num_rows) {
   if (feat.is_cuda()) {
     auto out = torch::empty({num_rows, feat.size(1)}, feat.options());
-    spmm_csr_hip(indptr, indices, feat, dst_scale, src_scale, out);
+    spmm_csr_hip(indptr, indices, feat, dst_scale, src_scale, row_order,
+                 out);
     return out;
   }
   if (feat.scalar_type() == torch::kBFloat16) {

@@ -49,7 +49,8 @@ torch::Tensor spmm_cpu(torch::Tensor indptr, torch::Tensor indices,
 // All tensors on device. feat fp32 [num_src, F], out fp32 [num_rows, F].
 void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
                   torch::Tensor feat, torch::Tensor dst_scale,
-                  torch::Tensor src_scale, torch::Tensor out);
+                  torch::Tensor src_scale, torch::Tensor row_order,
+                  torch::Tensor out);
 
 // out[i,:] = src[idx[i],:]
 void gather_rows_hip(torch::Tensor src, torch::Tensor idx, torch::Tensor out);

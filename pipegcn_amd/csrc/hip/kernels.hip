@@ -79,6 +79,7 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
                                 const T* __restrict__ feat,
                                 const float* __restrict__ dst_scale,
                                 const float* __restrict__ src_scale,
+                                const int32_t* __restrict__ row_order,
                                 T* __restrict__ out, int64_t num_rows,
                                 int64_t F, int64_t nchunks) {
   const int64_t wave_global =
@@ -97,6 +98,9 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
       r = pair / nchunks;
       chunk = pair % nchunks;
     }
+    // LPT scheduling: rows pre-sorted by degree descending — the heavy
+    // (lognormal-tail) rows start first, light rows backfill
+    if (row_order) r = row_order[r];
     const int64_t f0 = chunk * (kWave * VEC) + lane * VEC;
     if (f0 >= F) continue;
     const bool full = (f0 + VEC <= F);
@@ -182,9 +186,9 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
 template <typename T, int VEC>
 void launch_spmm(const int64_t* indptr, const int32_t* indices,
                  const T* feat, const float* dst_scale,
-                 const float* src_scale, T* out, int64_t num_src,
-                 int64_t num_rows, int64_t F, bool chunk_outer,
-                 hipStream_t stream) {
+                 const float* src_scale, const int32_t* row_order, T* out,
+                 int64_t num_src, int64_t num_rows, int64_t F,
+                 bool chunk_outer, hipStream_t stream) {
   const int64_t nchunks = (F + kWave * VEC - 1) / (kWave * VEC);
   const int threads = 256;  // 4 waves
   const int64_t npairs = num_rows * nchunks;
@@ -196,8 +200,8 @@ void launch_spmm(const int64_t* indptr, const int32_t* indices,
   if (blocks == 0) blocks = 1;
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, stream, indptr,
-                       indices, feat, dst_scale, src_scale, out, num_rows, F,
-                       nchunks);
+                       indices, feat, dst_scale, src_scale, row_order, out,
+                       num_rows, F, nchunks);
   };
   if (chunk_outer) {
     if (src_scale)
@@ -325,37 +329,24 @@ __global__ void colsum_partial_kernel(const float* __restrict__ x,
 }
 
 int pick_vec(int64_t F, int64_t num_rows, int64_t elem_size = 4) {
-  // Prefer the widest load whose column panel (num_src × 64·VEC × 4 B) stays
-  // L3-resident (256 MiB Infinity Cache; target ≤ ~160 MiB leaves room for
-  // the streaming index array). Overridable for A/B benchmarking.
-  // Measured on MI355X (profiles/README.md): the gather path tops out at
-  // ~6.5-7 TB/s logical regardless of L3 panel residency. What matters is
-  // (row,chunk)-pair count: with heavy-tailed degrees, few pairs leave the
-  // tail imbalanced (Reddit 233k rows: VEC1 with 4x the pairs beats VEC4 by
-  // 8%), while with plenty of pairs the widest load wins (ogbn-products
-  // 2.45M rows: VEC4 beats VEC1 by 17%). Rule: widest VEC whose pair count
-  // reaches ~40 per resident wave (16384 waves -> 640k pairs), else widest.
+  // With LPT row scheduling handling load balance, width is a pure
+  // throughput choice. Measured on MI355X (profiles/README.md): 8 B lane
+  // loads win at F=256 fp32 (VEC2 15.4 ms vs VEC1 15.7 / VEC4 16.0) and
+  // F=256 bf16 (VEC4); 16 B is within ~3%; 4 B loses up to 40% for bf16.
+  // Preference: 8 B, then 16 B, then 4 B — first width dividing F.
   if (const char* e = std::getenv("PIPEGCN_SPMM_VEC")) {
     int v = std::atoi(e);
     if ((v == 8 || v == 4 || v == 2 || v == 1) && F % v == 0 &&
         v * elem_size <= 16)
       return v;
   }
-  constexpr int64_t kTargetPairs = 640 * 1024;
-  int widest = 0;
-  // candidates from 16 B/lane down to 1 element
-  // lane loads below 4 B are never worth it (bf16 VEC1 measured 40% slower
-  // than VEC4 at F=256)
-  std::vector<int> cand;
-  for (int bytes = 16; bytes >= std::max<int64_t>(elem_size, 4); bytes >>= 1)
-    cand.push_back(bytes / static_cast<int>(elem_size));
-  for (int v : cand) {
-    if (F % v != 0) continue;
-    if (!widest) widest = v;
-    const int64_t nchunks = (F + 64 * v - 1) / (64 * v);
-    if (num_rows * nchunks >= kTargetPairs) return v;
+  (void)num_rows;
+  for (int bytes : {8, 16, 4}) {
+    if (bytes < elem_size) continue;
+    const int v = bytes / static_cast<int>(elem_size);
+    if (F % v == 0) return v;
   }
-  return widest;
+  return 1;
 }
 
 }  // namespace
@@ -363,34 +354,36 @@ int pick_vec(int64_t F, int64_t num_rows, int64_t elem_size = 4) {
 template <typename T>
 void spmm_dispatch(torch::Tensor& indptr, torch::Tensor& indices,
                    torch::Tensor& feat, const float* dsp, const float* ssp,
-                   torch::Tensor& out, int64_t num_src, int64_t num_rows,
-                   int64_t F, int vec, bool chunk_outer, hipStream_t stream) {
+                   const int32_t* rop, torch::Tensor& out, int64_t num_src,
+                   int64_t num_rows, int64_t F, int vec, bool chunk_outer,
+                   hipStream_t stream) {
   const T* fp = reinterpret_cast<const T*>(feat.data_ptr());
   T* op = reinterpret_cast<T*>(out.data_ptr());
   const int64_t* ip = indptr.data_ptr<int64_t>();
   const int32_t* xp = indices.data_ptr<int32_t>();
   switch (vec) {
     case 8:
-      launch_spmm<T, 8>(ip, xp, fp, dsp, ssp, op, num_src, num_rows, F,
+      launch_spmm<T, 8>(ip, xp, fp, dsp, ssp, rop, op, num_src, num_rows, F,
                         chunk_outer, stream);
       break;
     case 4:
-      launch_spmm<T, 4>(ip, xp, fp, dsp, ssp, op, num_src, num_rows, F,
+      launch_spmm<T, 4>(ip, xp, fp, dsp, ssp, rop, op, num_src, num_rows, F,
                         chunk_outer, stream);
       break;
     case 2:
-      launch_spmm<T, 2>(ip, xp, fp, dsp, ssp, op, num_src, num_rows, F,
+      launch_spmm<T, 2>(ip, xp, fp, dsp, ssp, rop, op, num_src, num_rows, F,
                         chunk_outer, stream);
       break;
     default:
-      launch_spmm<T, 1>(ip, xp, fp, dsp, ssp, op, num_src, num_rows, F,
+      launch_spmm<T, 1>(ip, xp, fp, dsp, ssp, rop, op, num_src, num_rows, F,
                         chunk_outer, stream);
   }
 }
 
 void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
                   torch::Tensor feat, torch::Tensor dst_scale,
-                  torch::Tensor src_scale, torch::Tensor out) {
+                  torch::Tensor src_scale, torch::Tensor row_order,
+                  torch::Tensor out) {
   TORCH_CHECK(feat.is_cuda() && out.is_cuda(), "spmm_csr_hip: device tensors");
   const bool bf16 = feat.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(bf16 || feat.scalar_type() == torch::kFloat,
@@ -415,17 +408,23 @@ void spmm_csr_hip(torch::Tensor indptr, torch::Tensor indices,
     TORCH_CHECK(src_scale.scalar_type() == torch::kFloat);
     ssp = src_scale.data_ptr<float>();
   }
+  const int32_t* rop = nullptr;
+  if (row_order.defined() && row_order.numel() > 0) {
+    TORCH_CHECK(row_order.is_contiguous() && row_order.numel() == num_rows &&
+                row_order.scalar_type() == torch::kInt);
+    rop = row_order.data_ptr<int32_t>();
+  }
   auto stream = current_stream();
   const int vec = pick_vec(F, num_rows, bf16 ? 2 : 4);
   bool chunk_outer = false;  // measured: chunk-inner wins at every shape
   if (const char* e = std::getenv("PIPEGCN_SPMM_ORDER"))
     chunk_outer = (e[0] == 'o');
   if (bf16)
-    spmm_dispatch<__hip_bfloat16>(indptr, indices, feat, dsp, ssp, out,
+    spmm_dispatch<__hip_bfloat16>(indptr, indices, feat, dsp, ssp, rop, out,
                                   num_src, num_rows, F, vec, chunk_outer,
                                   stream);
   else
-    spmm_dispatch<float>(indptr, indices, feat, dsp, ssp, out, num_src,
+    spmm_dispatch<float>(indptr, indices, feat, dsp, ssp, rop, out, num_src,
                          num_rows, F, vec, chunk_outer, stream);
 }
 

@@ -24,23 +24,33 @@ from pipegcn_amd import native
 
 @dataclass
 class CSR:
-    """CSR adjacency over rows; indices are column ids."""
+    """CSR adjacency over rows; indices are column ids.
+
+    row_order is the LPT schedule (rows sorted by degree descending): the
+    SpMM kernel processes heavy lognormal-tail rows first so light rows
+    backfill the wavefront slots.
+    """
 
     indptr: torch.Tensor  # int64 [num_rows + 1]
     indices: torch.Tensor  # int32 [nnz]
     num_rows: int
     num_cols: int
+    row_order: Optional[torch.Tensor] = None  # int32 [num_rows]
 
     @staticmethod
     def from_coo(u: torch.Tensor, v: torch.Tensor, num_rows: int,
                  num_cols: int) -> "CSR":
         """Build CSR with rows = v (dst), cols = u (src)."""
         indptr, indices = native().build_csr(u.cpu(), v.cpu(), num_rows)
-        return CSR(indptr, indices, num_rows, num_cols)
+        deg = indptr[1:] - indptr[:-1]
+        order = torch.argsort(deg, descending=True).to(torch.int32)
+        return CSR(indptr, indices, num_rows, num_cols, order)
 
     def to(self, device) -> "CSR":
         return CSR(self.indptr.to(device), self.indices.to(device),
-                   self.num_rows, self.num_cols)
+                   self.num_rows, self.num_cols,
+                   self.row_order.to(device)
+                   if self.row_order is not None else None)
 
     @property
     def nnz(self) -> int:

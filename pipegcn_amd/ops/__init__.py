@@ -24,8 +24,9 @@ def spmm(csr: CSR, feat: torch.Tensor,
     """out[r,:] = scale[r] * sum_{u in N(r)} src_scale[u] * feat[u,:]."""
     s = scale if scale is not None else torch.Tensor()
     ss = src_scale if src_scale is not None else torch.Tensor()
+    ro = csr.row_order if csr.row_order is not None else torch.Tensor()
     return native().spmm(csr.indptr, csr.indices, feat.contiguous(), s, ss,
-                         csr.num_rows)
+                         ro, csr.num_rows)
 
 
 class _SpmmMean(torch.autograd.Function):

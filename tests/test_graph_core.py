@@ -37,7 +37,7 @@ def test_spmm_cpu_matches_torch():
         feat = torch.randn(40, f)
         scale = torch.rand(25) + 0.5
         csr = CSR.from_coo(u, v, 25, 40)
-        out = native().spmm(csr.indptr, csr.indices, feat, scale, torch.Tensor(), 25)
+        out = native().spmm(csr.indptr, csr.indices, feat, scale, torch.Tensor(), torch.Tensor(), 25)
         ref = spmm_torch_ref(u, v, feat, 25, scale)
         assert torch.allclose(out, ref, atol=1e-5), f"F={f}"
 
@@ -46,7 +46,7 @@ def test_spmm_no_scale():
     u, v = random_coo(40, 25, 300)
     feat = torch.randn(40, 8)
     csr = CSR.from_coo(u, v, 25, 40)
-    out = native().spmm(csr.indptr, csr.indices, feat, torch.Tensor(), torch.Tensor(), 25)
+    out = native().spmm(csr.indptr, csr.indices, feat, torch.Tensor(), torch.Tensor(), torch.Tensor(), 25)
     assert torch.allclose(out, spmm_torch_ref(u, v, feat, 25), atol=1e-5)
 
 
@@ -85,9 +85,10 @@ def test_halo_graph_csc_is_transpose():
     feat = torch.randn(35, 5)
     gout = torch.randn(20, 5)
     # <A x, g> == <x, A^T g>
-    ax = native().spmm(g.csr.indptr, g.csr.indices, feat, torch.Tensor(), torch.Tensor(), 20)
+    ax = native().spmm(g.csr.indptr, g.csr.indices, feat, torch.Tensor(),
+                       torch.Tensor(), torch.Tensor(), 20)
     atg = native().spmm(g.csc.indptr, g.csc.indices, gout, torch.Tensor(),
-                        torch.Tensor(), 35)
+                        torch.Tensor(), torch.Tensor(), 35)
     assert torch.allclose((ax * gout).sum(), (feat * atg).sum(), atol=1e-3)
 
 
