@@ -340,8 +340,13 @@ int pick_vec(int64_t F, int64_t num_rows, int64_t elem_size = 4) {
         v * elem_size <= 16)
       return v;
   }
-  (void)num_rows;
-  for (int bytes : {8, 16, 4}) {
+  // row count flips the 8B/16B preference: 2.45M-row graphs (products)
+  // measured 16B 6% faster at F=256, 233k-row (reddit) 8B 4% faster.
+  const bool big = num_rows >= (1 << 20);
+  const std::initializer_list<int> pref =
+      big ? std::initializer_list<int>{16, 8, 4}
+          : std::initializer_list<int>{8, 16, 4};
+  for (int bytes : pref) {
     if (bytes < elem_size) continue;
     const int v = bytes / static_cast<int>(elem_size);
     if (F % v == 0) return v;
