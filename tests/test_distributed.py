@@ -25,7 +25,8 @@ def make_args(**kw):
                 n_feat=0, n_class=0, n_train=0, enable_pipeline=False,
                 feat_corr=False, grad_corr=False, corr_momentum=0.95,
                 use_pp=False, inductive=False, seed=5, log_every=100,
-                backend="gloo", eval=False, graph_name="test")
+                backend="gloo", eval=False, graph_name="test",
+                dtype="fp32")
     base.update(kw)
     return types.SimpleNamespace(**base)
 
@@ -260,7 +261,8 @@ def test_sync_batchnorm():
 # -------------------------------------------------- training equivalence
 
 
-def _equiv_worker(rank, world, tmpdir, pipeline, model="graphsage"):
+def _equiv_worker(rank, world, tmpdir, pipeline, model="graphsage",
+                  dtype="fp32"):
     from pipegcn_amd import trainer
     from pipegcn_amd.graph import partition
     from pipegcn_amd.parallel import context as ctx
@@ -273,7 +275,7 @@ def _equiv_worker(rank, world, tmpdir, pipeline, model="graphsage"):
     (u, v, n, ndata), part = _prepare_partitions(
         os.path.join(tmpdir, f"p{world}"), world)
     args = make_args(n_partitions=world,
-                     enable_pipeline=pipeline, model=model)
+                     enable_pipeline=pipeline, model=model, dtype=dtype)
     from pipegcn_amd.graph.datasets import data_stats
     args.n_feat, args.n_class, args.n_train = data_stats(ndata)
     s = trainer.run(part, args, device="cpu")
@@ -381,3 +383,22 @@ def test_chain_partition_empty_boundary(tmp_path):
     empties = run_distributed(_chain_worker, 3, args=(str(tmp_path),))
     # the chain's two end partitions each see one empty peer
     assert sum(empties) >= 2, empties
+
+
+def _bf16_worker(rank, world, tmpdir):
+    return _equiv_worker(rank, world, tmpdir, True, model="graphsage",
+                         dtype="bf16")
+
+
+def test_bf16_pipelined_training_converges(tmp_path):
+    """Optional bf16 compute: pipelined 2-rank training converges and the
+    bf16 losses track the fp32 run to a few percent."""
+    import functools
+
+    bf = run_distributed(_bf16_worker, 2, args=(str(tmp_path),))[0]
+    fp = run_distributed(
+        functools.partial(_equiv_worker, pipeline=True), 2,
+        args=(str(tmp_path),))[0]
+    assert bf[-1] < bf[0]
+    for a, b in zip(fp, bf):
+        assert abs(a - b) / abs(a) < 0.05, (fp, bf)
