@@ -59,7 +59,7 @@ class GCNLayer(nn.Module):
             ah = _SpmmSym.apply(graph, feat,
                                 inv_sqrt_all[: graph.num_in].contiguous(),
                                 inv_sqrt_all)
-            return self.linear(ah)
+            return ops.linear(ah, self.linear)
         assert isinstance(graph, FullGraph) and deg is None
         d = torch.rsqrt(graph.in_degrees().clamp(min=1.0)).contiguous()
         ah = ops.spmm(graph.csr, feat, d, src_scale=d)

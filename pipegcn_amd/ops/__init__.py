@@ -134,3 +134,31 @@ def sage_dual_linear(x1, x2, lin1, lin2):
     """
     return _SageDualLinear.apply(x1, x2, lin1.weight, lin2.weight, lin1.bias,
                                  lin2.bias)
+
+
+class _LinearColsum(torch.autograd.Function):
+    """nn.Linear forward via rocBLAS with a colsum bias grad — torch's
+    Linear backward picks a 1024-thread reduce for thin odd N (18.7 ms at
+    [2.45M,47]; profiles/README.md)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        out = torch.mm(x, w.t())
+        if b is not None:
+            out.add_(b)
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        x, w = ctx.saved_tensors
+        g = g.contiguous()
+        gb = native().colsum(g).to(w.dtype)
+        return g @ w, g.t() @ x, gb
+
+
+def linear(x, lin):
+    """lin(x) with the fast bias-grad path on GPU."""
+    if not x.is_cuda or lin.bias is None:
+        return lin(x)
+    return _LinearColsum.apply(x, lin.weight, lin.bias)
