@@ -108,7 +108,7 @@ class GCN(GNNBase):
                 h = self.layers[i](g, h, in_deg)
             else:
                 h = self.dropout(h)
-                h = self.layers[i](h)
+                h = ops.linear(h, self.layers[i])
             if i < self.n_layers - 1:
                 if self.use_norm:
                     h = self.norm[i](h)
