@@ -200,3 +200,25 @@ def test_ema_bf16_fallback():
     ref = (0.9 * avg.float() + 0.1 * x.float()).to(torch.bfloat16)
     ops.ema_update(avg, x, 0.9)
     assert torch.allclose(avg.float(), ref.float(), atol=0.05)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("n", [47, 100, 256])
+def test_linear_colsum_gpu(n):
+    """ops.linear (colsum bias grad) vs torch nn.Linear fwd+bwd."""
+    torch.manual_seed(n)
+    x = torch.randn(5000, 64, device="cuda", requires_grad=True)
+    lin = torch.nn.Linear(64, n).cuda()
+    out = ops.linear(x, lin)
+    ref = lin(x)
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+    g = torch.randn_like(out)
+    out.backward(g)
+    gx, gw, gb = x.grad.clone(), lin.weight.grad.clone(), \
+        lin.bias.grad.clone()
+    x.grad = None
+    lin.weight.grad = lin.bias.grad = None
+    lin(x).backward(g)
+    assert torch.allclose(gx, x.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(gw, lin.weight.grad, atol=1e-2, rtol=1e-3)
+    assert torch.allclose(gb, lin.bias.grad, atol=1e-2, rtol=1e-3)
