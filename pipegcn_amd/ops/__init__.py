@@ -97,6 +97,7 @@ class _SageDualLinear(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x1, x2, w1, w2, b1, b2):
         ctx.save_for_backward(x1, x2, w1, w2)
+        ctx.has_bias = b1 is not None
         if (w1.size(0) < 64 or not x1.is_cuda
                 or x1.dtype != torch.float32):
             # thin-N / CPU: the 128x128 MFMA tile would waste most of the
@@ -119,6 +120,8 @@ class _SageDualLinear(torch.autograd.Function):
         gx2 = g @ w2
         gw1 = g.t() @ x1
         gw2 = g.t() @ x2
+        if not ctx.has_bias:
+            return gx1, gx2, gw1, gw2, None, None
         gb = native().colsum(g)  # two-phase column sum (fastest measured)
         gb = gb.to(w1.dtype)
         return gx1, gx2, gw1, gw2, gb, gb
