@@ -113,3 +113,14 @@ def test_main_bf16(tmp_path):
     out = run_main(tmp_path, ["--dtype", "bf16", "--no-eval",
                               "--enable-pipeline"])
     assert "Epoch" in out and "nan" not in out
+
+
+def test_partition_tool(tmp_path):
+    r = subprocess.run(
+        [sys.executable, "-m", "pipegcn_amd.tools.partition_tool",
+         "--dataset", "synth-tiny", "--n-partitions", "2"],
+        cwd=tmp_path, env=dict(os.environ, PYTHONPATH=REPO),
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    assert (tmp_path / "partitions" / "synth-tiny-2-metis-vol-trans" /
+            "meta.json").exists()
