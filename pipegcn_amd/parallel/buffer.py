@@ -38,7 +38,7 @@ import os
 import queue
 import threading
 import time
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist

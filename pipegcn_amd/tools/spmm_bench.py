@@ -8,7 +8,6 @@ Usage (on a GPU box):
     python -m pipegcn_amd.tools.spmm_bench [--f 602 256] [--rounds 5]
 """
 import argparse
-import itertools
 import os
 import time
 
