@@ -322,6 +322,10 @@ class Buffer:
             t0 = self._stats_begin()
             with torch.cuda.stream(self._comm_stream):
                 self._feat_exchange(layer, feat, tag)
+                # the gather reads `feat` on the comm stream, but its
+                # storage is released by autograd on the compute stream —
+                # tell the caching allocator not to reuse it early
+                feat.record_stream(self._comm_stream)
             done_stream = self._comm_stream
             if self._corr_feat:
                 self._corr_stream.wait_stream(self._comm_stream)
