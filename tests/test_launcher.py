@@ -124,3 +124,14 @@ def test_partition_tool(tmp_path):
     assert r.returncode == 0, r.stderr
     assert (tmp_path / "partitions" / "synth-tiny-2-metis-vol-trans" /
             "meta.json").exists()
+
+
+def test_skip_partition_with_hints(tmp_path):
+    """--skip-partition reuses an existing partition dir; with
+    --n-feat/--n-class/--n-train hints it must not reload the dataset
+    (reference main.py:25-31)."""
+    run_main(tmp_path, ["--no-eval"])  # creates partitions/
+    out = run_main(tmp_path, ["--no-eval", "--skip-partition",
+                              "--n-feat", "16", "--n-class", "4",
+                              "--n-train", "79"])
+    assert "Epoch" in out
