@@ -60,6 +60,10 @@ if __name__ == "__main__":
 
     print(args)
 
+    if args.backend == "nccl" and not torch.cuda.is_available():
+        raise SystemExit("--backend nccl (RCCL) needs GPUs; "
+                         "use --backend gloo on CPU")
+
     if args.backend in ("gloo", "nccl"):
         if "CUDA_VISIBLE_DEVICES" in os.environ:
             devices = os.environ["CUDA_VISIBLE_DEVICES"].split(",")

@@ -339,6 +339,9 @@ def run(part: PartData, args, device: str = "cpu",
                           result_file_name))
 
     ctx.buffer.synchronize()
+    if device != "cpu":
+        print("Process {:03d} | peak GPU memory {:.2f} GB".format(
+            rank, torch.cuda.max_memory_allocated() / 2**30))
     summary = {
         "rank": rank,
         "losses": loss_hist,
@@ -378,6 +381,9 @@ def init_processes(rank, size, args, device: Optional[str] = None):
     if device is None:
         device = "cuda:0" if (args.backend == "nccl"
                               and torch.cuda.is_available()) else "cpu"
+    if args.backend == "nccl" and not torch.cuda.is_available():
+        raise RuntimeError(
+            "--backend nccl (RCCL) needs a GPU; use --backend gloo on CPU")
     if device.startswith("cuda"):
         torch.cuda.set_device(torch.device(device))
     graph_dir = os.path.join("partitions", args.graph_name)
