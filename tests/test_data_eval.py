@@ -144,3 +144,20 @@ def test_yelp_reference_layout(tmp_path, monkeypatch):
     # features are StandardScaler-normalized over train rows
     tr = ndata["feat"][ndata["train_mask"]]
     assert tr.mean(0).abs().max() < 0.5
+
+
+def test_npz_multilabel(tmp_path, monkeypatch):
+    n, f, c = 30, 4, 3
+    rng = np.random.RandomState(1)
+    os.makedirs(tmp_path / "dataset")
+    np.savez(tmp_path / "dataset" / "ml.npz",
+             src=rng.randint(0, n, 80), dst=rng.randint(0, n, 80),
+             feat=rng.randn(n, f).astype("f4"),
+             label=rng.randint(0, 2, (n, c)).astype("f4"),
+             train_mask=np.ones(n, bool), val_mask=np.zeros(n, bool),
+             test_mask=np.zeros(n, bool))
+    monkeypatch.chdir(tmp_path)
+    u, v, nn_, ndata = datasets.load_data("ml")
+    assert ndata["label"].dtype == torch.float32
+    n_feat, n_class, n_train = datasets.data_stats(ndata)
+    assert n_class == c and n_train == n
