@@ -17,12 +17,33 @@
 
 #include "common.h"
 
+#include <ATen/Parallel.h>
+
 #include <algorithm>
+#include <chrono>
+#include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <deque>
 #include <numeric>
 #include <random>
 #include <vector>
+
+namespace {
+struct PhaseTimer {
+  bool on = std::getenv("PIPEGCN_PART_VERBOSE") != nullptr;
+  std::chrono::steady_clock::time_point t =
+      std::chrono::steady_clock::now();
+  void lap(const char* name, int64_t n = -1) {
+    if (!on) return;
+    auto now = std::chrono::steady_clock::now();
+    double ms = std::chrono::duration<double, std::milli>(now - t).count();
+    fprintf(stderr, "[partition] %-18s %8.0f ms (n=%lld)\n", name, ms,
+            (long long)n);
+    t = now;
+  }
+};
+}  // namespace
 
 namespace {
 
@@ -74,6 +95,10 @@ int64_t heavy_edge_matching(const Graph& g, std::mt19937_64& rng,
 }
 
 Graph coarsen(const Graph& g, const std::vector<int32_t>& cmap, int64_t nc) {
+  // Deterministic parallel two-pass construction (the dominant cost at
+  // large scale): pass 1 counts each coarse node's distinct neighbors,
+  // pass 2 fills preallocated arrays — per-chunk scratch mark arrays, no
+  // locks, output independent of the thread count.
   Graph c;
   c.n = nc;
   c.nw.assign(nc, 0);
@@ -88,33 +113,51 @@ Graph coarsen(const Graph& g, const std::vector<int32_t>& cmap, int64_t nc) {
       m2[cu] = static_cast<int32_t>(u);
   }
   c.indptr.assign(nc + 1, 0);
-  std::vector<int32_t> mark(nc, -1);
-  std::vector<int64_t> slot(nc, 0);
-  // two passes: count, then fill
-  std::vector<int32_t> tmp_idx;
-  std::vector<int32_t> tmp_w;
-  for (int64_t cu = 0; cu < nc; ++cu) {
-    tmp_idx.clear();
-    tmp_w.clear();
-    for (int32_t u : {m1[cu], m2[cu]}) {
-      if (u < 0) continue;
-      for (int64_t e = g.indptr[u]; e < g.indptr[u + 1]; ++e) {
-        const int32_t cv = cmap[g.indices[e]];
-        if (cv == cu) continue;
-        if (mark[cv] != static_cast<int32_t>(cu)) {
-          mark[cv] = static_cast<int32_t>(cu);
-          slot[cv] = tmp_idx.size();
-          tmp_idx.push_back(cv);
-          tmp_w.push_back(g.ew[e]);
-        } else {
-          tmp_w[slot[cv]] += g.ew[e];
+  const int64_t grain = std::max<int64_t>(4096, nc / 64);
+  at::parallel_for(0, nc, grain, [&](int64_t b, int64_t en) {
+    std::vector<int32_t> mark(nc, -1);
+    for (int64_t cu = b; cu < en; ++cu) {
+      int64_t deg = 0;
+      for (int32_t u : {m1[cu], m2[cu]}) {
+        if (u < 0) continue;
+        for (int64_t e = g.indptr[u]; e < g.indptr[u + 1]; ++e) {
+          const int32_t cv = cmap[g.indices[e]];
+          if (cv == cu) continue;
+          if (mark[cv] != static_cast<int32_t>(cu)) {
+            mark[cv] = static_cast<int32_t>(cu);
+            deg++;
+          }
+        }
+      }
+      c.indptr[cu + 1] = deg;
+    }
+  });
+  for (int64_t i = 0; i < nc; ++i) c.indptr[i + 1] += c.indptr[i];
+  c.indices.resize(c.indptr[nc]);
+  c.ew.resize(c.indptr[nc]);
+  at::parallel_for(0, nc, grain, [&](int64_t b, int64_t en) {
+    std::vector<int32_t> mark(nc, -1);
+    std::vector<int64_t> slot(nc, 0);
+    for (int64_t cu = b; cu < en; ++cu) {
+      int64_t w = c.indptr[cu];
+      for (int32_t u : {m1[cu], m2[cu]}) {
+        if (u < 0) continue;
+        for (int64_t e = g.indptr[u]; e < g.indptr[u + 1]; ++e) {
+          const int32_t cv = cmap[g.indices[e]];
+          if (cv == cu) continue;
+          if (mark[cv] != static_cast<int32_t>(cu)) {
+            mark[cv] = static_cast<int32_t>(cu);
+            slot[cv] = w;
+            c.indices[w] = cv;
+            c.ew[w] = g.ew[e];
+            w++;
+          } else {
+            c.ew[slot[cv]] += g.ew[e];
+          }
         }
       }
     }
-    c.indptr[cu + 1] = c.indptr[cu] + static_cast<int64_t>(tmp_idx.size());
-    c.indices.insert(c.indices.end(), tmp_idx.begin(), tmp_idx.end());
-    c.ew.insert(c.ew.end(), tmp_w.begin(), tmp_w.end());
-  }
+  });
   return c;
 }
 
@@ -284,6 +327,8 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
     g0.nw.assign(N, 1);
   }
   std::vector<std::vector<int32_t>> cmaps;
+  PhaseTimer pt;
+  pt.lap("level0 copy", N);
 
   // --- coarsen
   const int64_t coarse_target = std::max<int64_t>(128 * nparts, 4096);
@@ -293,6 +338,7 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
     if (nc > levels.back().n * 95 / 100) break;  // matching stalled
     levels.push_back(coarsen(levels.back(), cmap, nc));
     cmaps.push_back(std::move(cmap));
+    pt.lap("match+coarsen", nc);
   }
 
   // --- initial partition on the coarsest level
@@ -328,6 +374,8 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
     }
   }
 
+  pt.lap("coarsest solve", levels.back().n);
+
   // --- uncoarsen + refine each level
   for (int64_t lvl = static_cast<int64_t>(levels.size()) - 2; lvl >= 0;
        --lvl) {
@@ -339,6 +387,7 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
         lvl == 0 ? std::max<int64_t>(n_refine_passes / 2, 2)
                  : std::max<int64_t>(n_refine_passes, 32);
     refine(levels[lvl], nparts, objective, lo_w, cap_w, passes, part);
+    pt.lap("project+refine", levels[lvl].n);
   }
 
   std::memcpy(out.data_ptr<int32_t>(), part.data(), N * sizeof(int32_t));
