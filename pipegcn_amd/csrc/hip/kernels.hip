@@ -10,8 +10,10 @@
 //    vector loads so each wave issues 1 KiB coalesced reads of a neighbor row.
 //  - SpMM is a bandwidth/gather-bound op: one wave owns one (dst-row, feature
 //    chunk); the edge loop is unrolled 4x to keep 4 gathers in flight per
-//    lane, and occupancy stays high (tiny VGPR footprint) so TLP hides HBM
-//    latency. The degree-divide epilogue is fused (scale argument).
+//    lane, rows are walked in descending-degree order (LPT scheduling — see
+//    pick_vec/profiles), and occupancy stays high (tiny VGPR footprint) so
+//    TLP hides HBM latency. The degree-divide epilogue is fused (scale
+//    argument); kernels are dtype-templated (fp32, bf16-with-fp32-accumulate).
 //  - The backward (transpose) SpMM is the same kernel run over the CSC of the
 //    halo graph, built once at setup — no atomics anywhere on the hot path.
 

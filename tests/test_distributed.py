@@ -402,3 +402,37 @@ def test_bf16_pipelined_training_converges(tmp_path):
     assert bf[-1] < bf[0]
     for a, b in zip(fp, bf):
         assert abs(a - b) / abs(a) < 0.05, (fp, bf)
+
+
+def _vanilla_corr_worker(rank, world, tmpdir):
+    """Corrections in NON-pipelined mode: the EMA tracks the FRESH value of
+    each epoch (no staleness), consumed instead of the raw recv."""
+    from pipegcn_amd.graph.halo import build_runtime_partition
+    from pipegcn_amd.parallel.buffer import Buffer
+
+    _, part = _prepare_partitions(tmpdir, world)
+    rp = build_runtime_partition(part)
+    F, m = 3, 0.5
+    buf = Buffer()
+    buf.init_buffer(rp.num_in, rp.num_all, rp.boundary, rp.recv_shape,
+                    [F], pipeline=False, backend="gloo", corr_feat=True,
+                    corr_momentum=m)
+    peer = 1 - rank
+    nhalo = rp.recv_shape[peer]
+    avg = 0.0
+    for epoch in range(3):
+        feat = torch.full((rp.num_in, F), float(epoch + 1 + 10 * rank))
+        h = buf.update(0, feat)
+        # synchronous mode: this epoch's value is already in the EMA
+        avg = m * avg + (1 - m) * float(epoch + 1 + 10 * peer)
+        halo = h[rp.num_in:rp.num_in + nhalo]
+        assert torch.allclose(halo, torch.full((nhalo, F), avg),
+                              atol=1e-5), f"epoch {epoch}"
+        buf.next_epoch()
+        comm_timer.clear()
+    buf.shutdown()
+    return True
+
+
+def test_vanilla_mode_feat_correction(tmp_path):
+    run_distributed(_vanilla_corr_worker, WORLD, args=(str(tmp_path),))
