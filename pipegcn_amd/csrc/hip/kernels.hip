@@ -185,6 +185,116 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
   }
 }
 
+// Multi-chunk variant (MEASURED NEGATIVE — kept env-gated as a recorded
+// experiment, PIPEGCN_SPMM_NC=8): ONE wave walks a row's edge list once and
+// covers up to NC column chunks, reading the index list once and turning
+// each edge's gathers into NC back-to-back requests on one contiguous
+// feature row. Measured 2.0-2.2x SLOWER than the chunk-per-wave default
+// (F=602: 41->80 ms; F=256: 15.5->31.5 ms, with 4-edge index lookahead):
+// concentrating the same gather instructions into NC-fold fewer waves
+// trades away wave-level parallelism, which on MI355X hides gather latency
+// far better than intra-wave ILP does.
+template <typename T, int VEC, int NC, bool HAS_SRC_SCALE>
+__global__ void spmm_csr_mc_kernel(const int64_t* __restrict__ indptr,
+                                   const int32_t* __restrict__ indices,
+                                   const T* __restrict__ feat,
+                                   const float* __restrict__ dst_scale,
+                                   const float* __restrict__ src_scale,
+                                   const int32_t* __restrict__ row_order,
+                                   T* __restrict__ out, int64_t num_rows,
+                                   int64_t F, int64_t ngroups) {
+  const int64_t wave_global =
+      (static_cast<int64_t>(blockIdx.x) * blockDim.x + threadIdx.x) / kWave;
+  const int lane = threadIdx.x & (kWave - 1);
+  const int64_t nwaves =
+      (static_cast<int64_t>(gridDim.x) * blockDim.x) / kWave;
+  const int64_t npairs = num_rows * ngroups;
+
+  for (int64_t pair = wave_global; pair < npairs; pair += nwaves) {
+    int64_t r = pair / ngroups;
+    const int64_t grp = pair % ngroups;
+    if (row_order) r = row_order[r];
+    const int64_t base =
+        grp * NC * (kWave * VEC) + static_cast<int64_t>(lane) * VEC;
+    if (base >= F) continue;
+
+    float acc[NC][VEC];
+#pragma unroll
+    for (int c = 0; c < NC; ++c)
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) acc[c][k] = 0.f;
+
+    const int64_t e_end = indptr[r + 1];
+    int64_t e = indptr[r];
+    auto body = [&](int64_t u) {
+      const float s = HAS_SRC_SCALE ? src_scale[u] : 1.f;
+      const T* frow = feat + u * F;
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+        const int64_t f = base + c * (kWave * VEC);
+        if (f + VEC <= F) {
+#pragma unroll
+          for (int k = 0; k < VEC; ++k)
+            acc[c][k] += s * to_f32(frow[f + k]);
+        } else if (f < F) {
+          for (int k = 0; f + k < F; ++k)
+            acc[c][k] += s * to_f32(frow[f + k]);
+        }
+      }
+    };
+    // 4-edge lookahead: one dwordx4 index load feeds 4 edges of gathers so
+    // the index-load latency is off the per-edge critical path
+    for (; e < e_end && (e & 3); ++e)
+      body(__builtin_nontemporal_load(indices + e));
+    for (; e + 4 <= e_end; e += 4) {
+      const int32x4 uu = __builtin_nontemporal_load(
+          reinterpret_cast<const int32x4*>(indices + e));
+      body(uu[0]);
+      body(uu[1]);
+      body(uu[2]);
+      body(uu[3]);
+    }
+    for (; e < e_end; ++e)
+      body(__builtin_nontemporal_load(indices + e));
+    const float s = dst_scale ? dst_scale[r] : 1.f;
+    T* orow = out + r * F;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      const int64_t f = base + c * (kWave * VEC);
+      if (f + VEC <= F) {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) st_nt(orow + f + k, acc[c][k] * s);
+      } else if (f < F) {
+        for (int k = 0; f + k < F; ++k) st(orow + f + k, acc[c][k] * s);
+      }
+    }
+  }
+}
+
+template <typename T, int VEC>
+void launch_spmm_mc(const int64_t* indptr, const int32_t* indices,
+                    const T* feat, const float* dst_scale,
+                    const float* src_scale, const int32_t* row_order, T* out,
+                    int64_t num_rows, int64_t F, hipStream_t stream) {
+  constexpr int NC = 8;
+  const int64_t nchunks = (F + kWave * VEC - 1) / (kWave * VEC);
+  const int64_t ngroups = (nchunks + NC - 1) / NC;
+  const int threads = 256;
+  int64_t blocks = (num_rows * ngroups * kWave + threads - 1) / threads;
+  blocks = std::min<int64_t>(blocks, 8 * 65536);
+  if (blocks == 0) blocks = 1;
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, stream, indptr,
+                       indices, feat, dst_scale, src_scale, row_order, out,
+                       num_rows, F, ngroups);
+  };
+  if (src_scale)
+    launch(HIP_KERNEL_NAME(spmm_csr_mc_kernel<T, VEC, NC, true>));
+  else
+    launch(HIP_KERNEL_NAME(spmm_csr_mc_kernel<T, VEC, NC, false>));
+  HIP_CHECK(hipGetLastError());
+}
+
 template <typename T, int VEC>
 void launch_spmm(const int64_t* indptr, const int32_t* indices,
                  const T* feat, const float* dst_scale,
@@ -368,6 +478,28 @@ void spmm_dispatch(torch::Tensor& indptr, torch::Tensor& indices,
   T* op = reinterpret_cast<T*>(out.data_ptr());
   const int64_t* ip = indptr.data_ptr<int64_t>();
   const int32_t* xp = indices.data_ptr<int32_t>();
+  if (const char* e = std::getenv("PIPEGCN_SPMM_NC")) {
+    if (std::atoi(e) == 8) {  // experimental whole-row multi-chunk variant
+      switch (vec) {
+        case 8:
+          launch_spmm_mc<T, 8>(ip, xp, fp, dsp, ssp, rop, op, num_rows, F,
+                               stream);
+          return;
+        case 4:
+          launch_spmm_mc<T, 4>(ip, xp, fp, dsp, ssp, rop, op, num_rows, F,
+                               stream);
+          return;
+        case 2:
+          launch_spmm_mc<T, 2>(ip, xp, fp, dsp, ssp, rop, op, num_rows, F,
+                               stream);
+          return;
+        default:
+          launch_spmm_mc<T, 1>(ip, xp, fp, dsp, ssp, rop, op, num_rows, F,
+                               stream);
+          return;
+      }
+    }
+  }
   switch (vec) {
     case 8:
       launch_spmm<T, 8>(ip, xp, fp, dsp, ssp, rop, op, num_src, num_rows, F,
