@@ -109,3 +109,24 @@ def test_partitioner_deterministic():
     c = native().partition_graph(indptr, indices, 4, 1, 0.05, 8, 8)
     # different seed may legitimately differ (not asserted equal)
     assert c.shape == a.shape
+
+
+def test_csr_row_order_is_lpt():
+    u, v = random_coo(60, 40, 500, seed=4)
+    csr = CSR.from_coo(u, v, 40, 60)
+    deg = (csr.indptr[1:] - csr.indptr[:-1])
+    ordered = deg[csr.row_order.long()]
+    assert (ordered[:-1] >= ordered[1:]).all(), \
+        "row_order must sort degrees descending"
+    assert sorted(csr.row_order.tolist()) == list(range(40))
+
+
+def test_synth_partition_invariants():
+    from pipegcn_amd.graph.synthetic import synth_partition
+
+    parts = [synth_partition("small", r, 3, seed=2) for r in range(3)]
+    assert sum(p.num_in for p in parts) == 1000
+    for p in parts:
+        assert int(p.ndata["in_degree"].sum()) == p.edges.shape[1]
+        assert p.edges[1].max() < p.num_in  # dsts are inner
+        assert p.edges[0].max() < p.num_local
