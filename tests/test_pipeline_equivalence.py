@@ -43,9 +43,12 @@ def _renumber(part):
     return new_id, ndata, u, v, nt
 
 
-def _simulate(parts, n_epochs, hidden, lr, n_train_global, seed):
+def _simulate(parts, n_epochs, hidden, lr, n_train_global, seed,
+              corr=False, momentum=0.8):
     """Sequential reference implementation of pipelined 2-partition
-    training (GraphSAGE, dropout 0, no norm, 2 conv layers)."""
+    training (GraphSAGE, dropout 0, no norm, 2 conv layers); with
+    corr=True the EMA smoothing correction replaces the raw stale values
+    on both directions (reference feature_buffer.py:186-191,137-140)."""
     import torch.nn.functional as F
 
     from pipegcn_amd.graph.csr import HaloGraph
@@ -97,6 +100,12 @@ def _simulate(parts, n_epochs, hidden, lr, n_train_global, seed):
     stale_grad = [[{j: torch.zeros(boundary[i][j].numel(), sizes[l])
                     for j in range(world) if j != i}
                    for l in range(L)] for i in range(world)]
+    avg_feat = [[{j: torch.zeros(boundary[j][i].numel(), sizes[l])
+                  for j in range(world) if j != i}
+                 for l in range(L)] for i in range(world)]
+    avg_grad = [[{j: torch.zeros(boundary[i][j].numel(), sizes[l])
+                  for j in range(world) if j != i}
+                 for l in range(L)] for i in range(world)]
     # stale_feat[i][l][j]: values rank i RECEIVED from j (j's boundary rows)
     # stale_grad[i][l][j]: grads rank i RECEIVED from j (for i's boundary
     #                      rows listed in boundary[i][j])
@@ -116,18 +125,20 @@ def _simulate(parts, n_epochs, hidden, lr, n_train_global, seed):
                 for j in range(world):
                     if j != i:
                         next_feat[j][l][i] = h[boundary[i][j]].detach()
+                src_f = avg_feat if corr else stale_feat
                 h_all = torch.cat(
-                    [h] + [stale_feat[i][l][j] for j in range(world)
+                    [h] + [src_f[i][l][j] for j in range(world)
                            if j != i])
                 if l > 0 and h_all.requires_grad:
                     def make_hook(i=i, l=l):
                         def hook(grad):
                             g2 = grad.clone()
+                            src_g = avg_grad if corr else stale_grad
                             for j in range(world):
                                 if j == i:
                                     continue
                                 # inject stale grads into my boundary rows
-                                g2[boundary[i][j]] += stale_grad[i][l][j]
+                                g2[boundary[i][j]] += src_g[i][l][j]
                                 # record halo-slice grads to send to j
                                 a, b = pl[i][j]
                                 next_grad[j][l][i] = grad[a:b].detach()
@@ -148,6 +159,22 @@ def _simulate(parts, n_epochs, hidden, lr, n_train_global, seed):
             p.grad /= n_train_global
         opt.step()
         losses.append(total.item())
+        if corr:
+            # the EMA updates when this epoch's transfer ARRIVES; it is
+            # consumed next epoch
+            for i in range(world):
+                for l in range(L):
+                    for j in range(world):
+                        if j == i:
+                            continue
+                        avg_feat[i][l][j] = (momentum * avg_feat[i][l][j]
+                                             + (1 - momentum)
+                                             * next_feat[i][l][j])
+                        ng = next_grad[i][l].get(
+                            j, torch.zeros(boundary[i][j].numel(),
+                                           sizes[l]))
+                        avg_grad[i][l][j] = (momentum * avg_grad[i][l][j]
+                                             + (1 - momentum) * ng)
         stale_feat, stale_grad = next_feat, next_grad
         # re-wrap dict layout: next_* entries keyed by sender
         stale_feat = [[{j: next_feat[i][l][j] for j in range(world)
@@ -159,7 +186,7 @@ def _simulate(parts, n_epochs, hidden, lr, n_train_global, seed):
     return losses
 
 
-def _dist_worker(rank, world, tmpdir):
+def _dist_worker(rank, world, tmpdir, corr=False):
     from pipegcn_amd import trainer
     from pipegcn_amd.graph.datasets import data_stats
     from pipegcn_amd.parallel import context as ctx
@@ -173,7 +200,8 @@ def _dist_worker(rank, world, tmpdir):
     ctx.reducer = Reducer()
     (u, v, n, ndata), part = _prepare_partitions(tmpdir + "/p", world)
     args = make_args(n_partitions=world, enable_pipeline=True, n_layers=2,
-                     norm="none", n_epochs=5, lr=0.05)
+                     norm="none", n_epochs=5, lr=0.05, feat_corr=corr,
+                     grad_corr=corr, corr_momentum=0.8)
     args.norm = None
     args.n_feat, args.n_class, args.n_train = data_stats(ndata)
     s = trainer.run(part, args, device="cpu")
@@ -192,6 +220,25 @@ def test_pipelined_matches_sequential_simulator(tmp_path):
              for r in range(WORLD)]
     sim_losses = _simulate(parts, 5, hidden=16, lr=0.05,
                            n_train_global=n_train, seed=5)
+    for a, b in zip(sim_losses, dist_losses):
+        assert abs(a - b) / max(abs(a), 1e-9) < 2e-3, \
+            (sim_losses, dist_losses)
+
+
+def _dist_worker_corr(rank, world, tmpdir):
+    return _dist_worker(rank, world, tmpdir, corr=True)
+
+
+def test_pipelined_corrections_match_simulator(tmp_path):
+    """Pipelined + feat/grad EMA correction vs the sequential simulator."""
+    from pipegcn_amd.graph import partition
+
+    dist_losses, n_train = run_distributed(
+        _dist_worker_corr, WORLD, args=(str(tmp_path),))[0]
+    parts = [partition.load_partition(str(tmp_path / "p"), r)
+             for r in range(WORLD)]
+    sim_losses = _simulate(parts, 5, hidden=16, lr=0.05,
+                           n_train_global=n_train, seed=5, corr=True)
     for a, b in zip(sim_losses, dist_losses):
         assert abs(a - b) / max(abs(a), 1e-9) < 2e-3, \
             (sim_losses, dist_losses)
