@@ -436,3 +436,40 @@ def _vanilla_corr_worker(rank, world, tmpdir):
 
 def test_vanilla_mode_feat_correction(tmp_path):
     run_distributed(_vanilla_corr_worker, WORLD, args=(str(tmp_path),))
+
+
+def _pp_worker(rank, world, tmpdir):
+    """--use-pp precompute: [feat ‖ mean_agg] must equal the global-graph
+    mean aggregation (reference train.py:169-189)."""
+    import types
+
+    from pipegcn_amd.graph.halo import build_runtime_partition
+    from pipegcn_amd.trainer import precompute
+
+    (u, v, n, ndata), part = _prepare_partitions(tmpdir, world)
+    rp = build_runtime_partition(part)
+    feat_pp = precompute(rp, types.SimpleNamespace(model="graphsage"))
+    F = rp.ndata["feat"].shape[1]
+    assert feat_pp.shape == (rp.num_in, 2 * F)
+
+    # global reference (same mapping as the halo-agg test)
+    from pipegcn_amd.graph.partition import assign_partitions
+    pvec = assign_partitions(u, v, n, world, "metis", "vol", 0).long()
+    order = torch.argsort(pvec, stable=True)
+    new_gid = torch.empty(n, dtype=torch.long)
+    new_gid[order] = torch.arange(n)
+    gfeat = torch.zeros(n, F)
+    gfeat[new_gid] = ndata["feat"]
+    deg = torch.bincount(new_gid[v], minlength=n).float().clamp(min=1)
+    ref = torch.zeros(n, F)
+    ref.index_add_(0, new_gid[v], gfeat[new_gid[u]])
+    ref /= deg.unsqueeze(1)
+    gid = rp.ndata["gid"]
+    assert torch.allclose(feat_pp[:, :F], rp.ndata["feat"], atol=1e-6)
+    assert torch.allclose(feat_pp[:, F:], ref[gid], atol=1e-4), \
+        (feat_pp[:, F:] - ref[gid]).abs().max()
+    return True
+
+
+def test_use_pp_precompute_matches_global(tmp_path):
+    run_distributed(_pp_worker, WORLD, args=(str(tmp_path),))
