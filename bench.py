@@ -162,6 +162,9 @@ def main():
         optimizer.step()
         wait_s += comm_timer.tot_time()
         comm_timer.clear()
+    # drain the pipelined buffer queue so the final epoch's boundary
+    # transfers are inside the timed region (they belong to the epoch)
+    ctx.buffer.synchronize()
     barrier_sync()
     elapsed = time.time() - t0
     comm_busy = ctx.buffer.pop_comm_stats()

@@ -22,8 +22,8 @@ if __name__ == "__main__":
     args = create_parser()
     if args.fix_seed is False:
         if args.parts_per_node < args.n_partitions:
-            warnings.warn("Please enable `--fix-seed` for multi-node "
-                          "training.")
+            warnings.warn("multi-node runs need identical seeds on every "
+                          "node — pass --fix-seed")
         args.seed = random.randint(0, 1 << 31)
 
     if args.graph_name == "":
@@ -35,8 +35,9 @@ if __name__ == "__main__":
 
     if args.skip_partition:
         if args.n_feat == 0 or args.n_class == 0 or args.n_train == 0:
-            warnings.warn("Specifying `--n-feat`, `--n-class` and "
-                          "`--n-train` saves data loading time.")
+            warnings.warn("--skip-partition without --n-feat/--n-class/"
+                          "--n-train forces a full data load just to count "
+                          "them — pass all three to skip it")
             u, v, n, ndata = datasets.load_data(
                 args.dataset, nparts_hint=args.n_partitions, seed=args.seed)
             args.n_feat, args.n_class, args.n_train = datasets.data_stats(

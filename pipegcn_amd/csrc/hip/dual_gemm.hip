@@ -53,6 +53,10 @@ __global__ __launch_bounds__(256) void sage_dual_gemm_kernel(
   const int64_t m0 = static_cast<int64_t>(blockIdx.x) * BM;
   const int64_t n0 = static_cast<int64_t>(blockIdx.y) * BN;
   const bool interior = (m0 + BM <= M) && (n0 + BN <= N);
+  // float4 loads need 16B alignment: base ptrs are 16B-aligned (torch
+  // allocator) but row offsets row*K are only 4B-aligned when K%4!=0
+  // (K=602 for reddit) — use 4 scalar dword loads then (still guard-free)
+  const bool k16 = (K & 3) == 0;
 
   f32x16 acc[2][2] = {};
 
@@ -76,8 +80,12 @@ __global__ __launch_bounds__(256) void sage_dual_gemm_kernel(
 #pragma unroll
         for (int rr = 0; rr < BM / A_RSTEP; ++rr) {
           const int row = a_r + rr * A_RSTEP;
-          const float4 val = *reinterpret_cast<const float4*>(
-              X + (m0 + row) * K + k0 + a_c);
+          const float* p = X + (m0 + row) * K + k0 + a_c;
+          float4 val;
+          if (k16)
+            val = *reinterpret_cast<const float4*>(p);
+          else
+            val = make_float4(p[0], p[1], p[2], p[3]);
           a_lds[row][a_c + 0] = val.x;
           a_lds[row][a_c + 1] = val.y;
           a_lds[row][a_c + 2] = val.z;
@@ -103,8 +111,12 @@ __global__ __launch_bounds__(256) void sage_dual_gemm_kernel(
       if (interior && kfull) {
 #pragma unroll
         for (int q = 0; q < TBK / 2; q += 4) {
-          const float4 val = *reinterpret_cast<const float4*>(
-              W + (n0 + b_n) * K + k0 + b_k + q);
+          const float* p = W + (n0 + b_n) * K + k0 + b_k + q;
+          float4 val;
+          if (k16)
+            val = *reinterpret_cast<const float4*>(p);
+          else
+            val = make_float4(p[0], p[1], p[2], p[3]);
           b_lds[b_k + q + 0][b_n] = val.x;
           b_lds[b_k + q + 1][b_n] = val.y;
           b_lds[b_k + q + 2][b_n] = val.z;

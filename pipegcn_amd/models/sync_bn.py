@@ -6,6 +6,16 @@ computes dx = (w/n)/std * (n*g - dbias - x_hat*dweight). Collectives run on
 the main-thread process group (RCCL on GPU, gloo on CPU); the column
 reductions themselves are [N,F]->[F] torch ops (rocBLAS/eager — profiled as
 negligible next to SpMM).
+
+Divergences from the reference implementation (the forward/backward algebra
+itself must match numerically for parity, and is textbook sync-BN):
+ - statistics are accumulated in fp32 even under bf16 compute (the reference
+   is fp32-only);
+ - the column sums go through our native two-phase colsum kernel on GPU
+   (kernels.hip) instead of torch.sum — the thin-N eager reduction was an
+   18.7 ms pathology (profiles/README.md);
+ - explicit dtype casts at the normalize/return boundary so the module is
+   usable inside a bf16 model.
 """
 import torch
 import torch.distributed as dist

@@ -239,6 +239,9 @@ class Buffer:
     def _wait_host(self, evt: threading.Event):
         while not evt.wait(timeout=60.0):
             self._check_err()
+        # a failing comm thread sets the event to unblock waiters — check
+        # unconditionally so the error surfaces HERE, not an epoch later
+        self._check_err()
         evt.clear()
 
     def _wait_dev(self, evt):
@@ -383,6 +386,12 @@ class Buffer:
             with torch.cuda.stream(self._comm_stream):
                 self._transport.all_to_all(send, self._b_recv[layer],
                                            key=("b", layer), tag=tag)
+                # the sends read row slices of `grad` on the comm stream;
+                # autograd frees grad's storage on the compute stream right
+                # after the hook returns — same allocator-reuse hazard as
+                # the feat path (and ProcessGroupNCCL's internal
+                # recordStream is off under TORCH_NCCL_AVOID_RECORD_STREAMS)
+                grad.record_stream(self._comm_stream)
                 if _DEBUG:
                     self._debug_log("grad", tag, send, self._b_recv[layer])
             done_stream = self._comm_stream

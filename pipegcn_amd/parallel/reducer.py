@@ -31,7 +31,8 @@ class Reducer:
         self._group = None
 
     def init(self, model: torch.nn.Module, group=None):
-        self._params = [p for _, p in sorted(model.named_parameters())]
+        self._params = [p for _, p in sorted(model.named_parameters(),
+                                             key=lambda kv: kv[0])]
         n = sum(p.numel() for p in self._params)
         dev = self._params[0].device if self._params else "cpu"
         self._flat = torch.zeros(n, device=dev)

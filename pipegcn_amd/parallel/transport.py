@@ -128,18 +128,39 @@ def exchange_index_lists(wanted: List[Optional[torch.Tensor]],
     out: List[Optional[torch.Tensor]] = [None] * size
     use_cuda = dist.get_backend(group) == "nccl"
     dev = "cuda" if use_cuda else "cpu"
+    # phase 1: sizes — ONE batched group over all ring peers (unmatched
+    # singleton isend/recv is a hang-prone pattern on NCCL/RCCL; grouped
+    # P2P is what the data path uses too)
+    sends = {r: wanted[r].to(dev).contiguous()
+             for _, r in ring_peers(rank, size)}
+    n_recv = {l: torch.zeros(1, dtype=torch.long, device=dev)
+              for l, _ in ring_peers(rank, size)}
+    p2p = []
     for left, right in ring_peers(rank, size):
-        w = wanted[right].to(dev)
-        n_send = torch.tensor([w.numel()], dtype=torch.long, device=dev)
-        n_recv = torch.zeros(1, dtype=torch.long, device=dev)
-        req = dist.isend(n_send, dst=right)
-        dist.recv(n_recv, src=left)
-        req.wait()
-        buf = torch.zeros(int(n_recv.item()), dtype=torch.long, device=dev)
-        req = dist.isend(w, dst=right) if w.numel() > 0 else None
-        if buf.numel() > 0:
-            dist.recv(buf, src=left)
-        if req is not None:
-            req.wait()
-        out[left], _ = torch.sort(buf.cpu())
+        p2p.append(dist.P2POp(dist.irecv, n_recv[left], left, group))
+        p2p.append(dist.P2POp(dist.isend,
+                              torch.tensor([sends[right].numel()],
+                                           dtype=torch.long, device=dev),
+                              right, group))
+    for w in dist.batch_isend_irecv(p2p):
+        w.wait()
+    if use_cuda:
+        torch.cuda.current_stream().synchronize()
+    # phase 2: index lists — one batched group; zero-length entries are
+    # skipped consistently on both sides (sender knows numel, recver knows n)
+    bufs = {l: torch.zeros(int(n_recv[l].item()), dtype=torch.long,
+                           device=dev) for l in n_recv}
+    p2p = []
+    for left, right in ring_peers(rank, size):
+        if bufs[left].numel() > 0:
+            p2p.append(dist.P2POp(dist.irecv, bufs[left], left, group))
+        if sends[right].numel() > 0:
+            p2p.append(dist.P2POp(dist.isend, sends[right], right, group))
+    if p2p:
+        for w in dist.batch_isend_irecv(p2p):
+            w.wait()
+    if use_cuda:
+        torch.cuda.current_stream().synchronize()
+    for left in bufs:
+        out[left], _ = torch.sort(bufs[left].cpu())
     return out
