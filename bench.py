@@ -50,6 +50,12 @@ def main():
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--backend", type=str, default=None,
                     help="override torch.distributed backend (tests)")
+    ap.add_argument("--solo-of", type=int, default=0,
+                    help="sizing mode: run rank 0 of an N-way partition "
+                         "alone on one GPU (full buffer layout + event "
+                         "choreography, empty wire) — the papers100M "
+                         "288 GB/GPU check: "
+                         "`--shape ogbn-papers100m --solo-of 8`")
     args = ap.parse_args()
 
     env_world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -86,9 +92,12 @@ def main():
     # batch norm normalizes by the GLOBAL train count (reference semantics,
     # consistent only when every node is a train node — the inductive setup)
     train_frac = 1.0 if args.norm == "batch" else 0.66
-    part = synth_partition(args.shape, rank, world, seed=0,
-                           train_frac=train_frac)
-    rp = build_runtime_partition(part, device=device)
+    solo = args.solo_of > 1
+    if solo and world > 1:
+        raise SystemExit("--solo-of is a single-process sizing mode")
+    part = synth_partition(args.shape, rank, args.solo_of if solo else world,
+                           seed=0, train_frac=train_frac)
+    rp = build_runtime_partition(part, device=device, solo=solo)
 
     pipeline = not args.no_pipeline
     norm = None if args.norm == "none" else args.norm
@@ -103,7 +112,8 @@ def main():
                            corr_feat=args.feat_corr,
                            corr_grad=args.grad_corr,
                            device=device, group=comm_group,
-                           collect_stats=True, dtype=dtype)
+                           collect_stats=True, dtype=dtype,
+                           solo_world=args.solo_of)
 
     model = GraphSAGE(layer_size, F.relu, use_pp=args.use_pp, dropout=0.5,
                       norm=norm, n_linear=0,
@@ -198,7 +208,7 @@ def main():
             # a baseline ratio — reduced precision must not inflate it
             "vs_baseline": (epoch_s / baseline
                             if args.shape == "reddit"
-                            and args.dtype == "fp32" else None),
+                            and args.dtype == "fp32" and not solo else None),
             "dtype": args.dtype,
             "data": "synthetic",
             "config": {
@@ -211,7 +221,11 @@ def main():
                 "pipeline": pipeline,
                 "use_pp": args.use_pp,
                 "norm": args.norm,
-                "parallelism": f"graph-partition dp{world}",
+                "parallelism": (f"solo rank0-of-{args.solo_of} (sizing)"
+                                if solo else
+                                f"graph-partition dp{world}"),
+                "num_halo": rp.num_all - rp.num_in,
+                "num_edges": rp.graph.csr.nnz,
                 "comm_busy_s_per_epoch": comm_busy / args.steps,
                 "comm_wait_s_per_epoch": wait_s / args.steps,
                 "boundary_comm_overlap_pct": overlap_pct,

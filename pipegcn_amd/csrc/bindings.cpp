@@ -77,6 +77,52 @@ static torch::Tensor colsum(torch::Tensor x) {
   return x.sum(0);
 }
 
+// GPU-only fused ops (the Python wrappers fall back to eager torch on CPU)
+
+static std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p,
+                                              int64_t seed) {
+  TORCH_CHECK(x.is_cuda(), "fused dropout is the GPU path");
+  auto y = torch::empty_like(x);
+  auto mask = torch::empty({(x.numel() + 7) / 8},
+                           x.options().dtype(torch::kUInt8));
+  dropout_fwd_hip(x, y, mask, p, seed);
+  return {y, mask};
+}
+
+static torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask,
+                                 double p) {
+  auto dx = torch::empty_like(dy);
+  dropout_bwd_hip(dy.contiguous(), mask, dx, p);
+  return dx;
+}
+
+static std::vector<torch::Tensor> layer_norm_relu_fwd(torch::Tensor x,
+                                                      torch::Tensor w,
+                                                      torch::Tensor b,
+                                                      double eps, bool relu) {
+  TORCH_CHECK(x.is_cuda(), "fused layer_norm_relu is the GPU path");
+  auto y = torch::empty_like(x);
+  auto xhat = torch::empty_like(x);
+  auto rstd = torch::empty({x.size(0)}, x.options().dtype(torch::kFloat));
+  layer_norm_relu_fwd_hip(x.contiguous(), w, b, eps, relu, y, xhat, rstd);
+  return {y, xhat, rstd};
+}
+
+static std::vector<torch::Tensor> layer_norm_relu_bwd(
+    torch::Tensor dy, torch::Tensor xhat, torch::Tensor rstd,
+    torch::Tensor w, torch::Tensor b, bool relu) {
+  auto dx = torch::empty_like(dy);
+  const int64_t N = dy.size(0);
+  const int64_t F = dy.size(1);
+  const int64_t nblk = std::min<int64_t>(1024, (N + 3) / 4);
+  auto dw_part = torch::empty({nblk * 4, F},
+                              dy.options().dtype(torch::kFloat));
+  auto db_part = torch::empty_like(dw_part);
+  layer_norm_relu_bwd_hip(dy.contiguous(), xhat, rstd, w, b, relu, dx,
+                          dw_part, db_part);
+  return {dx, colsum_hip(dw_part), colsum_hip(db_part)};
+}
+
 static void ema_update(torch::Tensor avg, torch::Tensor x, double momentum) {
   if (avg.is_cuda()) {
     ema_update_hip(avg, x, momentum);
@@ -99,5 +145,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &colsum, "out[n] = sum_m x[m,n]");
   m.def("sage_dual_gemm", &sage_dual_gemm,
         "out = x1 @ w1^T + x2 @ w2^T + bias (MFMA fp32, fused)");
+  m.def("dropout_fwd", &dropout_fwd,
+        "fused dropout, bitpacked mask -> (y, mask)");
+  m.def("dropout_bwd", &dropout_bwd, "dx = mask ? dy/(1-p) : 0");
+  m.def("layer_norm_relu_fwd", &layer_norm_relu_fwd,
+        "fused LayerNorm[+ReLU] -> (y, xhat, rstd)");
+  m.def("layer_norm_relu_bwd", &layer_norm_relu_bwd,
+        "-> (dx, dweight, dbias)");
   m.attr("with_hip") = true;
 }

@@ -67,6 +67,25 @@ void ema_update_hip(torch::Tensor avg, torch::Tensor x, double momentum);
 // out[n] = sum_m x[m, n] — fast column reduction for bias/BN grads.
 torch::Tensor colsum_hip(torch::Tensor x);
 
+// Fused dropout with a bitpacked mask (1 bit/elem): y = mask ? x/(1-p) : 0.
+// mask uint8[(n+7)/8]; seed drives a counter-based RNG (reproducible).
+void dropout_fwd_hip(torch::Tensor x, torch::Tensor y, torch::Tensor mask,
+                     double p, int64_t seed);
+void dropout_bwd_hip(torch::Tensor dy, torch::Tensor mask, torch::Tensor dx,
+                     double p);
+
+// Fused LayerNorm [+ReLU]: y = [relu](w * xhat + b), xhat/rstd saved for
+// backward (wave-per-row, fp32 statistics, F <= 1024).
+void layer_norm_relu_fwd_hip(torch::Tensor x, torch::Tensor w,
+                             torch::Tensor b, double eps, bool relu,
+                             torch::Tensor y, torch::Tensor xhat,
+                             torch::Tensor rstd);
+// dw_part/db_part: fp32 [nwaves, F] per-wave partials (column-sum on host).
+void layer_norm_relu_bwd_hip(torch::Tensor dy, torch::Tensor xhat,
+                             torch::Tensor rstd, torch::Tensor w,
+                             torch::Tensor b, bool relu, torch::Tensor dx,
+                             torch::Tensor dw_part, torch::Tensor db_part);
+
 // Fused dual GEMM for the GraphSAGE layer epilogue:
 //   out[M,N] = x1[M,K] @ w1t[K,N] + x2[M,K] @ w2t[K,N] + b[N]
 // fp32, MFMA (v_mfma_f32_16x16x4_f32). Weights pre-transposed to [K,N].

@@ -40,14 +40,23 @@ class RuntimePartition:
         return self
 
 
-def build_runtime_partition(part: PartData,
-                            device: str = "cpu") -> RuntimePartition:
+def build_runtime_partition(part: PartData, device: str = "cpu",
+                            solo: bool = False) -> RuntimePartition:
     """Construct the training-time halo graph for this rank's partition.
 
     Requires torch.distributed to be initialized (world_size may be 1).
+
+    solo=True: single-process sizing mode — the partition is laid out as if
+    its full world (len(node_offsets)-1 ranks) were running, with the real
+    halo rows and recv shapes, but no peers exist: the boundary exchange is
+    skipped and boundary sets are empty (nothing will be sent). Used by
+    `bench.py --solo-of N` for the papers100M 288 GB/GPU sizing run.
     """
     rank = dist.get_rank() if dist.is_initialized() else 0
     size = dist.get_world_size() if dist.is_initialized() else 1
+    if solo:
+        assert size == 1 and rank == 0, "solo mode is single-process"
+        size = len(part.node_offsets) - 1
     num_in = part.num_in
     num_halo = part.halo_gnid.numel()
     num_all = num_in + num_halo
@@ -94,7 +103,10 @@ def build_runtime_partition(part: PartData,
         wanted[j] = part.halo_gnid[sel] - offsets[j]
 
     # --- boundary exchange (who needs my rows)
-    if size > 1:
+    if solo:
+        boundary = [torch.zeros(0, dtype=torch.long) for _ in range(size)]
+        boundary[rank] = None
+    elif size > 1:
         got = exchange_index_lists(wanted)
         boundary: List[Optional[torch.Tensor]] = [
             new_id[g] if g is not None else None for g in got

@@ -104,13 +104,11 @@ class GCN(GNNBase):
             if i < self.n_layers - self.n_linear:
                 if self.training:
                     h = ctx.buffer.update(i, h)
-                h = self.dropout(h)
+                h = self._drop(h)
                 h = self.layers[i](g, h, in_deg)
             else:
-                h = self.dropout(h)
+                h = self._drop(h)
                 h = ops.linear(h, self.layers[i])
             if i < self.n_layers - 1:
-                if self.use_norm:
-                    h = self.norm[i](h)
-                h = self.activation(h)
+                h = self._norm_act(i, h)
         return h

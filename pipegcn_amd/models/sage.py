@@ -84,6 +84,23 @@ class GNNBase(nn.Module):
             self.norm = nn.ModuleList()
         self.dropout = nn.Dropout(p=dropout)
 
+    def _drop(self, h):
+        """Dropout: fused bitmask kernel on GPU, eager elsewhere."""
+        if self.training and h.is_cuda and self.dropout.p > 0:
+            return ops.fused_dropout(h, self.dropout.p)
+        return self.dropout(h)
+
+    def _norm_act(self, i, h):
+        """Inter-layer norm + activation: fused LayerNorm+ReLU on GPU."""
+        if h.is_cuda and self.use_norm and isinstance(self.norm[i],
+                                                      nn.LayerNorm):
+            fuse_relu = self.activation is torch.nn.functional.relu
+            h = ops.layer_norm_relu(h, self.norm[i], relu=fuse_relu)
+            return h if fuse_relu else self.activation(h)
+        if self.use_norm:
+            h = self.norm[i](h)
+        return self.activation(h)
+
 
 class GraphSAGE(GNNBase):
     def __init__(self, layer_size, activation, use_pp, dropout=0.5,
@@ -114,13 +131,11 @@ class GraphSAGE(GNNBase):
             if i < self.n_layers - self.n_linear:
                 if self.training and (i > 0 or not self.use_pp):
                     h = ctx.buffer.update(i, h)
-                h = self.dropout(h)
+                h = self._drop(h)
                 h = self.layers[i](g, h, in_deg)
             else:
-                h = self.dropout(h)
+                h = self._drop(h)
                 h = ops.linear(h, self.layers[i])
             if i < self.n_layers - 1:
-                if self.use_norm:
-                    h = self.norm[i](h)
-                h = self.activation(h)
+                h = self._norm_act(i, h)
         return h

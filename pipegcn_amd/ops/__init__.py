@@ -139,6 +139,66 @@ def sage_dual_linear(x1, x2, lin1, lin2):
                                  lin2.bias)
 
 
+class _FusedDropout(torch.autograd.Function):
+    """Dropout with a BITPACKED mask (1 bit/element vs torch's byte mask —
+    8x less saved memory) and the scale fused into the same pass. The RNG is
+    counter-based (splitmix64 keyed on a host-drawn seed), so the op is
+    reproducible given torch.manual_seed. GPU path only."""
+
+    @staticmethod
+    def forward(ctx, x, p, seed):
+        y, mask = native().dropout_fwd(x.contiguous(), p, seed)
+        ctx.save_for_backward(mask)
+        ctx.p = p
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (mask,) = ctx.saved_tensors
+        return native().dropout_bwd(dy, mask, ctx.p), None, None
+
+
+def fused_dropout(x: torch.Tensor, p: float) -> torch.Tensor:
+    """Training-mode dropout on GPU (call only when training and p > 0)."""
+    # host-drawn 63-bit seed: deterministic under torch.manual_seed
+    seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+    return _FusedDropout.apply(x, p, seed)
+
+
+class _LayerNormReLU(torch.autograd.Function):
+    """Fused LayerNorm [+ReLU] (replaces the eager nn.LayerNorm + F.relu
+    pair between layers, /root/reference/module/model.py:53-56).
+
+    Saves xhat + rstd only — eager autograd keeps {LN input, ReLU output}
+    (two [N,F] tensors); this keeps one. Statistics in fp32 for both dtypes;
+    the ReLU mask is recomputed in backward from w*xhat+b.
+    """
+
+    @staticmethod
+    def forward(ctx, x, w, b, eps, relu):
+        y, xhat, rstd = native().layer_norm_relu_fwd(
+            x.contiguous(), w.contiguous(), b.contiguous(), eps, relu)
+        ctx.save_for_backward(xhat, rstd, w, b)
+        ctx.relu = relu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        xhat, rstd, w, b = ctx.saved_tensors
+        dx, dw, db = native().layer_norm_relu_bwd(dy, xhat, rstd, w, b,
+                                                  ctx.relu)
+        return dx, dw, db, None, None
+
+
+def layer_norm_relu(x: torch.Tensor, ln: torch.nn.LayerNorm,
+                    relu: bool) -> torch.Tensor:
+    """ln(x) [+ relu] through the fused gfx950 kernel (GPU path)."""
+    # fp32 weight views keep the kernel dtype-simple; the casts are autograd
+    # nodes, so bf16 params still receive their grads (no-op for fp32)
+    return _LayerNormReLU.apply(x, ln.weight.float(), ln.bias.float(),
+                                ln.eps, relu)
+
+
 class _LinearColsum(torch.autograd.Function):
     """nn.Linear forward via rocBLAS with a colsum bias grad — torch's
     Linear backward picks a 1024-thread reduce for thin odd N (18.7 ms at

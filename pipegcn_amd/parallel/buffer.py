@@ -63,9 +63,16 @@ class Buffer:
                     corr_feat: bool = False, corr_grad: bool = False,
                     corr_momentum: float = 0.95, device: str = "cpu",
                     group=None, collect_stats: bool = False,
-                    dtype: torch.dtype = torch.float32):
+                    dtype: torch.dtype = torch.float32,
+                    solo_world: int = 0):
         self._rank = dist.get_rank() if dist.is_initialized() else 0
         self._size = dist.get_world_size() if dist.is_initialized() else 1
+        if solo_world > 1:
+            # sizing mode (bench.py --solo-of N): full N-rank buffer layout,
+            # streams and event choreography, but no peers — transfers skip
+            # the wire and recv buffers stay zero (= epoch-0 semantics).
+            assert self._size == 1
+            self._size = solo_world
         self._num_in = num_in
         self._num_all = num_all
         self._boundary = boundary
@@ -80,7 +87,7 @@ class Buffer:
         self._device = torch.device(device)
         self._use_cuda = self._device.type == "cuda"
         self._transport = (RingTransport(group)
-                           if self._size > 1 else None)
+                           if self._size > 1 and solo_world <= 1 else None)
         self._collect_stats = collect_stats
         self._stats_pairs = []  # (start_evt, end_evt) on comm stream
         self._comm_busy_host = 0.0
@@ -359,8 +366,10 @@ class Buffer:
                 continue
             ops.gather_rows_into(feat.detach(), self._boundary[j],
                                  self._f_send[layer][j])
-        self._transport.all_to_all(self._f_send[layer], self._f_recv[layer],
-                                   key=("f", layer), tag=tag)
+        if self._transport is not None:
+            self._transport.all_to_all(self._f_send[layer],
+                                       self._f_recv[layer],
+                                       key=("f", layer), tag=tag)
         if _DEBUG:
             self._debug_log("feat", tag, self._f_send[layer],
                             self._f_recv[layer])
@@ -384,8 +393,9 @@ class Buffer:
             self._comm_stream.wait_event(self._b_consumed_evt[layer])
             t0 = self._stats_begin()
             with torch.cuda.stream(self._comm_stream):
-                self._transport.all_to_all(send, self._b_recv[layer],
-                                           key=("b", layer), tag=tag)
+                if self._transport is not None:
+                    self._transport.all_to_all(send, self._b_recv[layer],
+                                               key=("b", layer), tag=tag)
                 # the sends read row slices of `grad` on the comm stream;
                 # autograd frees grad's storage on the compute stream right
                 # after the hook returns — same allocator-reuse hazard as
@@ -408,8 +418,9 @@ class Buffer:
             self._stats_end(t0)
         else:
             t0 = self._stats_begin()
-            self._transport.all_to_all(send, self._b_recv[layer],
-                                       key=("b", layer), tag=tag)
+            if self._transport is not None:
+                self._transport.all_to_all(send, self._b_recv[layer],
+                                           key=("b", layer), tag=tag)
             if _DEBUG:
                 self._debug_log("grad", tag, send, self._b_recv[layer])
             if self._corr_grad:

@@ -23,6 +23,7 @@ ext = CUDAExtension(
         "pipegcn_amd/csrc/partitioner.cpp",
         "pipegcn_amd/csrc/hip/kernels.hip",
         "pipegcn_amd/csrc/hip/dual_gemm.hip",
+        "pipegcn_amd/csrc/hip/elementwise.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -222,3 +222,82 @@ def test_linear_colsum_gpu(n):
     assert torch.allclose(gx, x.grad, atol=1e-3, rtol=1e-3)
     assert torch.allclose(gw, lin.weight.grad, atol=1e-2, rtol=1e-3)
     assert torch.allclose(gb, lin.bias.grad, atol=1e-2, rtol=1e-3)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("f", [128, 256, 602, 100])
+@pytest.mark.parametrize("relu", [True, False])
+def test_layer_norm_relu_gpu(f, relu):
+    """Fused LayerNorm[+ReLU] fwd+bwd vs eager fp32 torch."""
+    torch.manual_seed(f)
+    n = 3111
+    x = torch.randn(n, f, device="cuda", requires_grad=True)
+    ln = torch.nn.LayerNorm(f).cuda()
+    with torch.no_grad():
+        ln.weight.uniform_(0.5, 1.5)
+        ln.bias.uniform_(-0.5, 0.5)
+    out = ops.layer_norm_relu(x, ln, relu=relu)
+    ref = ln(x)
+    if relu:
+        ref = torch.nn.functional.relu(ref)
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+    g = torch.randn_like(out)
+    out.backward(g)
+    gx, gw, gb = (x.grad.clone(), ln.weight.grad.clone(),
+                  ln.bias.grad.clone())
+    x.grad = None
+    ln.weight.grad = ln.bias.grad = None
+    r = ln(x)
+    if relu:
+        r = torch.nn.functional.relu(r)
+    r.backward(g)
+    assert torch.allclose(gx, x.grad, atol=1e-4, rtol=1e-3)
+    assert torch.allclose(gw, ln.weight.grad, atol=1e-2, rtol=1e-3)
+    assert torch.allclose(gb, ln.bias.grad, atol=1e-2, rtol=1e-3)
+
+
+@pytest.mark.gpu
+def test_layer_norm_relu_bf16_gpu():
+    """bf16 fused LN+ReLU tracks the fp32 eager reference."""
+    torch.manual_seed(0)
+    x32 = torch.randn(2048, 256, device="cuda")
+    x = x32.to(torch.bfloat16).requires_grad_(True)
+    ln = torch.nn.LayerNorm(256).cuda().to(torch.bfloat16)
+    out = ops.layer_norm_relu(x, ln, relu=True)
+    ref = torch.nn.functional.relu(
+        torch.nn.functional.layer_norm(
+            x.float(), (256,), ln.weight.float(), ln.bias.float(), ln.eps))
+    assert torch.allclose(out.float(), ref, atol=0.05, rtol=0.05)
+    out.sum().backward()
+    assert torch.isfinite(x.grad.float()).all()
+    assert torch.isfinite(ln.weight.grad.float()).all()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("p", [0.1, 0.5])
+def test_fused_dropout_gpu(p):
+    """Bitmask dropout: scale/zero pattern, statistics, bwd consistency."""
+    torch.manual_seed(1)
+    x = torch.randn(4000, 130, device="cuda", requires_grad=True)
+    y = ops.fused_dropout(x, p)
+    kept = y != 0
+    # kept elements are x/(1-p) (up to p's 1/65536 quantization)
+    assert torch.allclose(y[kept], x.detach()[kept] / (1 - p), rtol=1e-3)
+    frac = kept.float().mean().item()
+    assert abs(frac - (1 - p)) < 0.01, frac
+    # backward uses the SAME mask
+    g = torch.randn_like(y)
+    y.backward(g)
+    assert torch.allclose(x.grad[kept], g[kept] / (1 - p), rtol=1e-3)
+    assert (x.grad[~kept] == 0).all()
+
+
+@pytest.mark.gpu
+def test_fused_dropout_determinism_gpu():
+    """Same torch seed => same mask (counter-based RNG, host-drawn seed)."""
+    x = torch.randn(999, 67, device="cuda")
+    torch.manual_seed(42)
+    y1 = ops.fused_dropout(x, 0.5)
+    torch.manual_seed(42)
+    y2 = ops.fused_dropout(x, 0.5)
+    assert torch.equal(y1, y2)
