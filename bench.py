@@ -88,6 +88,14 @@ def main():
     import torch.nn.functional as F
     import types
 
+    import sys
+
+    def phase(msg):
+        if rank == 0:
+            print(f"[bench t={time.time() - t_start:.0f}s] {msg}",
+                  file=sys.stderr, flush=True)
+
+    t_start = time.time()
     torch.manual_seed(0)
     # batch norm normalizes by the GLOBAL train count (reference semantics,
     # consistent only when every node is a train node — the inductive setup)
@@ -97,7 +105,10 @@ def main():
         raise SystemExit("--solo-of is a single-process sizing mode")
     part = synth_partition(args.shape, rank, args.solo_of if solo else world,
                            seed=0, train_frac=train_frac)
+    phase(f"partition generated: {part.num_in} inner nodes, "
+          f"{part.edges.shape[1]} edges, {part.halo_gnid.numel()} halo")
     rp = build_runtime_partition(part, device=device, solo=solo)
+    phase("runtime partition built (CSR+CSC on device)")
 
     pipeline = not args.no_pipeline
     norm = None if args.norm == "none" else args.norm
@@ -133,6 +144,7 @@ def main():
     if args.use_pp:
         pp_args = types.SimpleNamespace(model="graphsage")
         feat = precompute(rp, pp_args)
+        phase("use-pp precompute done")
     in_deg = rp.ndata["in_degree"]
     labels = rp.ndata["label"][: rp.num_train]
     model.train()
@@ -154,8 +166,11 @@ def main():
         if device.startswith("cuda"):
             torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
+    for w in range(args.warmup):
         step()
+        phase(f"warmup {w + 1}/{args.warmup}" + (
+            f" (peak {torch.cuda.max_memory_allocated() / 2**30:.1f} GB)"
+            if device.startswith("cuda") else ""))
     if device.startswith("cuda"):
         torch.cuda.reset_peak_memory_stats()
     ctx.buffer.pop_comm_stats()  # reset overlap stats

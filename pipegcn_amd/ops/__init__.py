@@ -22,6 +22,11 @@ def spmm(csr: CSR, feat: torch.Tensor,
          scale: Optional[torch.Tensor] = None,
          src_scale: Optional[torch.Tensor] = None) -> torch.Tensor:
     """out[r,:] = scale[r] * sum_{u in N(r)} src_scale[u] * feat[u,:]."""
+    if feat.shape[0] < csr.num_cols:
+        raise ValueError(
+            f"spmm: feat has {feat.shape[0]} rows but the graph references "
+            f"{csr.num_cols} source nodes (an under-sized feat would be an "
+            "out-of-bounds GPU gather)")
     s = scale if scale is not None else torch.Tensor()
     ss = src_scale if src_scale is not None else torch.Tensor()
     ro = csr.row_order if csr.row_order is not None else torch.Tensor()

@@ -109,7 +109,7 @@ def exchange_halo_values(rp: RuntimePartition,
     size = dist.get_world_size() if dist.is_initialized() else 1
     rank = dist.get_rank() if dist.is_initialized() else 0
     if size == 1:
-        return values
+        return _pad_halo_zeros(rp, values.unsqueeze(1)).squeeze(1)
     from pipegcn_amd.parallel.transport import RingTransport
 
     send = [None] * size
@@ -123,6 +123,16 @@ def exchange_halo_values(rp: RuntimePartition,
     RingTransport().all_to_all(send, recv, key="halo_scalar", tag=2)
     return torch.cat([values] + [recv[j].squeeze(1) for j in range(size)
                                  if j != rank])
+
+
+def _pad_halo_zeros(rp: RuntimePartition, t: torch.Tensor) -> torch.Tensor:
+    """Extend [num_in, F] to [num_all, F] with zero halo rows (solo sizing
+    mode has real halo slots but no peers to fill them — epoch-0 semantics)."""
+    if t.shape[0] >= rp.num_all:
+        return t
+    pad = torch.zeros(rp.num_all - t.shape[0], *t.shape[1:],
+                      device=t.device, dtype=t.dtype)
+    return torch.cat([t, pad])
 
 
 def precompute(rp: RuntimePartition, args) -> torch.Tensor:
@@ -150,7 +160,10 @@ def precompute(rp: RuntimePartition, args) -> torch.Tensor:
         feat_all = torch.cat([feat] + [recv[j] for j in range(size)
                                        if j != rank])
     else:
-        feat_all = feat
+        # solo sizing mode: the halo graph references [0, num_all) source
+        # rows even with no peers — pad with zeros (an under-sized feat
+        # tensor would be an out-of-bounds SpMM gather)
+        feat_all = _pad_halo_zeros(rp, feat)
     inv_deg = (1.0 / rp.ndata["in_degree"].clamp(min=1.0)).contiguous()
     mean_feat = ops.spmm(rp.graph.csr, feat_all, inv_deg)
     return torch.cat([feat, mean_feat], dim=1)
