@@ -58,6 +58,13 @@ def main():
                          "`--shape ogbn-papers100m --solo-of 8`")
     args = ap.parse_args()
 
+    if args.solo_of > 1:
+        # sizing runs push toward the 288 GB HBM ceiling where caching-
+        # allocator fragmentation (tens of GB reserved-but-unallocated)
+        # turns a fitting run into an OOM; expandable segments remove it
+        os.environ.setdefault("PYTORCH_ALLOC_CONF",
+                              "expandable_segments:True")
+
     env_world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     world = max(env_world, 1)
