@@ -151,6 +151,10 @@ def main():
     if args.use_pp:
         pp_args = types.SimpleNamespace(model="graphsage")
         feat = precompute(rp, pp_args)
+        # bench never evals: the raw [num_all, F] feature tensor is dead
+        # once the precomputed [num_in, 2F] input exists (papers100M
+        # sizing: −14.1 GB of the 288 GB budget)
+        rp.ndata.pop("feat", None)
         phase("use-pp precompute done")
     in_deg = rp.ndata["in_degree"]
     labels = rp.ndata["label"][: rp.num_train]
