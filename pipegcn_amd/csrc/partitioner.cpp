@@ -8,10 +8,25 @@
 //   2. INITIAL PARTITION on the coarsest graph: farthest-point seeded,
 //      weight-balanced multi-source BFS growth.
 //   3. UNCOARSEN: project the assignment back up, boundary-refining at
-//      every level (FM-style single-node moves, edge-weight gains, node
-//      -weight balance; the 'vol' objective adds the node's replica-count
-//      delta — the communication-volume term of the reference's
-//      objtype='vol').
+//      every level (FM-style single-node moves; the 'vol' objective uses
+//      EXACT O(deg) communication-volume gains — the moving node's own
+//      replica delta plus every neighbor's replica add/remove, tracked
+//      with per-node neighbor-partition counters — whenever the counter
+//      table fits PIPEGCN_PART_EXACT_VOL_MB (default 4096 MB; 111M nodes
+//      x 8 parts = 3.4 GB), the reference's objtype='vol'; above the
+//      limit it falls back to the approximate own-delta-only gain).
+//
+// Scale notes (papers100M: 111M nodes, multi-billion edges):
+//   - level 0 dedups + strips self-loops into int32 arrays ONCE (the
+//     symmetrized input duplicates every edge; duplicates would also
+//     break the distinct-neighbor volume logic) and carries unit edge/
+//     node weights implicitly (no 2x int32/int64 arrays at full scale);
+//   - coarsening is a deterministic parallel two-pass; refinement and
+//     matching are serial sweeps (measured in profiles/).
+//
+// PIPEGCN_PART_CHECK_VOL=1: verify every accepted vol move against a
+// brute-force local volume recomputation (the exact-gain correctness
+// oracle; used by tests, debug only).
 //
 // Exported as partition_graph_cpu with the original signature.
 
@@ -43,16 +58,25 @@ struct PhaseTimer {
     t = now;
   }
 };
+
+int64_t env_int(const char* name, int64_t dflt) {
+  const char* s = std::getenv(name);
+  return s ? std::atoll(s) : dflt;
+}
 }  // namespace
 
 namespace {
 
+// CSR graph. Level 0 has implicit unit edge/node weights (ewp/nwp null)
+// to stay lean at 100M+ nodes; coarse levels own their weight arrays.
 struct Graph {
   int64_t n = 0;
   std::vector<int64_t> indptr;
   std::vector<int32_t> indices;
-  std::vector<int32_t> ew;  // edge weights
-  std::vector<int64_t> nw;  // node weights (fine nodes contained)
+  std::vector<int32_t> ew_s;  // edge weights (empty => unit)
+  std::vector<int64_t> nw_s;  // node weights (empty => unit)
+  int64_t ew(int64_t e) const { return ew_s.empty() ? 1 : ew_s[e]; }
+  int64_t nw(int64_t x) const { return nw_s.empty() ? 1 : nw_s[x]; }
 };
 
 // heavy-edge matching; returns coarse count and fine->coarse map
@@ -70,8 +94,8 @@ int64_t heavy_edge_matching(const Graph& g, std::mt19937_64& rng,
     for (int64_t e = g.indptr[u]; e < g.indptr[u + 1]; ++e) {
       const int32_t v = g.indices[e];
       if (v == u || match[v] >= 0) continue;
-      if (g.ew[e] > best_w) {
-        best_w = g.ew[e];
+      if (g.ew(e) > best_w) {
+        best_w = g.ew(e);
         best = v;
       }
     }
@@ -101,12 +125,12 @@ Graph coarsen(const Graph& g, const std::vector<int32_t>& cmap, int64_t nc) {
   // locks, output independent of the thread count.
   Graph c;
   c.n = nc;
-  c.nw.assign(nc, 0);
+  c.nw_s.assign(nc, 0);
   // members of each coarse node (at most 2)
   std::vector<int32_t> m1(nc, -1), m2(nc, -1);
   for (int64_t u = 0; u < g.n; ++u) {
     const int32_t cu = cmap[u];
-    c.nw[cu] += g.nw[u];
+    c.nw_s[cu] += g.nw(u);
     if (m1[cu] < 0)
       m1[cu] = static_cast<int32_t>(u);
     else
@@ -134,7 +158,7 @@ Graph coarsen(const Graph& g, const std::vector<int32_t>& cmap, int64_t nc) {
   });
   for (int64_t i = 0; i < nc; ++i) c.indptr[i + 1] += c.indptr[i];
   c.indices.resize(c.indptr[nc]);
-  c.ew.resize(c.indptr[nc]);
+  c.ew_s.resize(c.indptr[nc]);
   at::parallel_for(0, nc, grain, [&](int64_t b, int64_t en) {
     std::vector<int32_t> mark(nc, -1);
     std::vector<int64_t> slot(nc, 0);
@@ -149,10 +173,10 @@ Graph coarsen(const Graph& g, const std::vector<int32_t>& cmap, int64_t nc) {
             mark[cv] = static_cast<int32_t>(cu);
             slot[cv] = w;
             c.indices[w] = cv;
-            c.ew[w] = g.ew[e];
+            c.ew_s[w] = static_cast<int32_t>(g.ew(e));
             w++;
           } else {
-            c.ew[slot[cv]] += g.ew[e];
+            c.ew_s[slot[cv]] += static_cast<int32_t>(g.ew(e));
           }
         }
       }
@@ -206,7 +230,7 @@ void initial_partition(const Graph& g, int64_t nparts, int64_t cap_w,
     const int64_t s = seeds[k];
     if (part[s] < 0) {
       part[s] = static_cast<int32_t>(k);
-      psize[k] += g.nw[s];
+      psize[k] += g.nw(s);
       frontier[k].push_back(s);
     }
   }
@@ -222,7 +246,7 @@ void initial_partition(const Graph& g, int64_t nparts, int64_t cap_w,
           const int32_t y = g.indices[e];
           if (part[y] < 0) {
             part[y] = static_cast<int32_t>(k);
-            psize[k] += g.nw[y];
+            psize[k] += g.nw(y);
             frontier[k].push_back(y);
             claimed = true;
             progress = true;
@@ -241,56 +265,172 @@ void initial_partition(const Graph& g, int64_t nparts, int64_t cap_w,
       const int64_t k =
           std::min_element(psize.begin(), psize.end()) - psize.begin();
       part[x] = static_cast<int32_t>(k);
-      psize[k] += g.nw[x];
+      psize[k] += g.nw(x);
     }
   }
 }
 
-// FM-style boundary refinement (single-node moves, edge-weight gains)
+// brute-force volume term of one node: nw(v) * #{p != part[v] : some
+// neighbor of v is in p}  (the CHECK_VOL oracle)
+int64_t node_vol(const Graph& g, const std::vector<int32_t>& part,
+                 std::vector<int64_t>& scratch, int64_t v) {
+  int64_t reps = 0;
+  for (int64_t e = g.indptr[v]; e < g.indptr[v + 1]; ++e) {
+    const int32_t u = g.indices[e];
+    if (u == v) continue;
+    const int32_t p = part[u];
+    if (p != part[v] && scratch[p] == 0) {
+      scratch[p] = 1;
+      reps++;
+    }
+  }
+  for (int64_t e = g.indptr[v]; e < g.indptr[v + 1]; ++e)
+    scratch[part[g.indices[e]]] = 0;
+  return g.nw(v) * reps;
+}
+
+// FM-style boundary refinement (single-node moves).
+//  objective 0: weighted edge-cut gains.
+//  objective 1 ("vol"): exact communication-volume gains when the
+//    nbrcnt table fits (see header comment); approximate otherwise.
 void refine(const Graph& g, int64_t nparts, int64_t objective, int64_t lo_w,
             int64_t cap_w, int64_t passes, std::vector<int32_t>& part) {
   const int64_t n = g.n;
   std::vector<int64_t> psize(nparts, 0);
-  for (int64_t x = 0; x < n; ++x) psize[part[x]] += g.nw[x];
+  for (int64_t x = 0; x < n; ++x) psize[part[x]] += g.nw(x);
+
+  // exact-vol neighbor-partition counters: nbrcnt[v*nparts+p] = number of
+  // v's incident edges leading into partition p (level-0 graphs are
+  // deduped+loop-free, coarse graphs are simple => edge count == distinct
+  // neighbor count, which is what volume semantics need)
+  bool exact = false;
+  std::vector<int32_t> nbrcnt;
+  if (objective == 1) {
+    const int64_t limit_b =
+        env_int("PIPEGCN_PART_EXACT_VOL_MB", 4096) * 1024 * 1024;
+    if (n * nparts * (int64_t)sizeof(int32_t) <= limit_b) {
+      exact = true;
+      nbrcnt.assign(n * nparts, 0);
+      at::parallel_for(0, n, 8192, [&](int64_t b, int64_t en) {
+        for (int64_t x = b; x < en; ++x)
+          for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+            const int32_t v = g.indices[e];
+            if (v != x) nbrcnt[x * nparts + part[v]]++;
+          }
+      });
+    }
+  }
+  const bool check =
+      exact && std::getenv("PIPEGCN_PART_CHECK_VOL") != nullptr;
+  std::vector<int64_t> vol_scratch;
+  if (check) vol_scratch.assign(nparts, 0);
+
   std::vector<int64_t> cnt(nparts, 0);
+  std::vector<int64_t> addq(nparts, 0);
   std::vector<int32_t> touched;
   touched.reserve(64);
   for (int64_t pass = 0; pass < passes; ++pass) {
     int64_t moves = 0;
     for (int64_t x = 0; x < n; ++x) {
       const int32_t a = part[x];
-      if (psize[a] - g.nw[x] < lo_w) continue;
+      if (psize[a] - g.nw(x) < lo_w) continue;
       bool boundary = false;
       for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
-        const int32_t q = part[g.indices[e]];
+        const int32_t v = g.indices[e];
+        if (v == x) continue;  // self-loops carry no cut/volume
+        const int32_t q = part[v];
         if (cnt[q] == 0) touched.push_back(q);
-        cnt[q] += g.ew[e];
+        cnt[q] += g.ew(e);
         if (q != a) boundary = true;
       }
       if (boundary) {
-        int32_t best = a;
-        int64_t best_gain = 0;
-        for (const int32_t q : touched) {
-          if (q == a || psize[q] + g.nw[x] > cap_w) continue;
-          int64_t gain = cnt[q] - cnt[a];
-          if (objective == 1) {
-            // communication-volume term: replica-count delta of x
-            int64_t rep_a = 0, rep_q = 0;
-            for (const int32_t t : touched) {
-              if (t != a && cnt[t] > 0) rep_a++;
-              if (t != q && cnt[t] > 0) rep_q++;
-            }
-            gain += rep_a - rep_q;
+        int64_t rep_a = 0;  // replicas of x if it stays in a
+        if (objective == 1)
+          for (const int32_t t : touched)
+            if (t != a && cnt[t] > 0) rep_a++;
+        int64_t rem_a = 0;  // nbr volume freed when x leaves a
+        if (exact) {
+          for (const int32_t t : touched) addq[t] = 0;
+          for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+            const int32_t v = g.indices[e];
+            if (v == x) continue;
+            const int32_t b = part[v];
+            const int32_t* row = nbrcnt.data() + (int64_t)v * nparts;
+            if (b != a && row[a] == 1) rem_a += g.nw(v);
+            for (const int32_t t : touched)
+              if (t != b && row[t] == 0) addq[t] += g.nw(v);
           }
-          if (gain > best_gain) {
-            best_gain = gain;
+        }
+        int32_t best = a;
+        int64_t best_gain = 0, best_vol = 0, best_cut = 0;
+        for (const int32_t q : touched) {
+          if (q == a || psize[q] + g.nw(x) > cap_w) continue;
+          const int64_t cut_gain = cnt[q] - cnt[a];
+          if (objective == 1) {
+            int64_t rep_q = 0;
+            for (const int32_t t : touched)
+              if (t != q && cnt[t] > 0) rep_q++;
+            if (exact) {
+              // exact volume gain: own replicas + neighbor add/remove
+              const int64_t vol_gain =
+                  g.nw(x) * (rep_a - rep_q) + rem_a - addq[q];
+              // lexicographic (volume, cut): volume is the objective,
+              // cut breaks ties; only strictly-improving moves
+              // (vol>0, or vol-neutral with cut>0) are ever accepted
+              if ((vol_gain > best_vol ||
+                   (vol_gain == best_vol && cut_gain > best_cut)) &&
+                  (vol_gain > 0 || (vol_gain == 0 && cut_gain > 0))) {
+                best = q;
+                best_vol = vol_gain;
+                best_cut = cut_gain;
+              }
+              continue;
+            }
+            // approximate: own replica delta only
+            const int64_t gain = cut_gain + rep_a - rep_q;
+            if (gain > best_gain) {
+              best_gain = gain;
+              best = q;
+            }
+            continue;
+          }
+          if (cut_gain > best_gain) {
+            best_gain = cut_gain;
             best = q;
           }
         }
         if (best != a) {
+          if (check) {
+            // oracle: recompute local volume of {x} u N(x) before/after
+            int64_t before = node_vol(g, part, vol_scratch, x);
+            for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e)
+              if (g.indices[e] != x)
+                before += node_vol(g, part, vol_scratch, g.indices[e]);
+            part[x] = best;
+            int64_t after = node_vol(g, part, vol_scratch, x);
+            for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e)
+              if (g.indices[e] != x)
+                after += node_vol(g, part, vol_scratch, g.indices[e]);
+            part[x] = a;
+            // duplicate neighbors in S would double-count; level-0 is
+            // deduped and coarse graphs are simple, so each neighbor
+            // appears once
+            TORCH_CHECK(before - after == best_vol,
+                        "exact vol gain mismatch: predicted ", best_vol,
+                        " actual ", before - after, " at node ", x);
+          }
+          if (exact) {
+            // maintain neighbor counters for the move a -> best
+            for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+              const int32_t v = g.indices[e];
+              if (v == x) continue;
+              nbrcnt[(int64_t)v * nparts + a]--;
+              nbrcnt[(int64_t)v * nparts + best]++;
+            }
+          }
           part[x] = best;
-          psize[a] -= g.nw[x];
-          psize[best] += g.nw[x];
+          psize[a] -= g.nw(x);
+          psize[best] += g.nw(x);
           moves++;
         }
       }
@@ -314,21 +454,51 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
 
   std::mt19937_64 rng(seed);
 
-  // level 0 = the input graph, unit weights
+  // level 0: dedup + strip self-loops ONCE (the symmetrized input carries
+  // every undirected edge twice; duplicate edges would break the
+  // distinct-neighbor volume counters), unit weights implicit.
   std::vector<Graph> levels(1);
   {
     Graph& g0 = levels[0];
     g0.n = N;
     const int64_t* ip = indptr.data_ptr<int64_t>();
     const int32_t* xp = indices.data_ptr<int32_t>();
-    g0.indptr.assign(ip, ip + N + 1);
-    g0.indices.assign(xp, xp + g0.indptr[N]);
-    g0.ew.assign(g0.indices.size(), 1);
-    g0.nw.assign(N, 1);
+    g0.indptr.assign(N + 1, 0);
+    at::parallel_for(0, N, 8192, [&](int64_t b, int64_t en) {
+      std::vector<int32_t> row;
+      for (int64_t x = b; x < en; ++x) {
+        row.assign(xp + ip[x], xp + ip[x + 1]);
+        std::sort(row.begin(), row.end());
+        int64_t deg = 0;
+        int32_t prev = -1;
+        for (const int32_t v : row)
+          if (v != x && v != prev) {
+            prev = v;
+            deg++;
+          }
+        g0.indptr[x + 1] = deg;
+      }
+    });
+    for (int64_t i = 0; i < N; ++i) g0.indptr[i + 1] += g0.indptr[i];
+    g0.indices.resize(g0.indptr[N]);
+    at::parallel_for(0, N, 8192, [&](int64_t b, int64_t en) {
+      std::vector<int32_t> row;
+      for (int64_t x = b; x < en; ++x) {
+        row.assign(xp + ip[x], xp + ip[x + 1]);
+        std::sort(row.begin(), row.end());
+        int64_t w = g0.indptr[x];
+        int32_t prev = -1;
+        for (const int32_t v : row)
+          if (v != x && v != prev) {
+            prev = v;
+            g0.indices[w++] = v;
+          }
+      }
+    });
   }
   std::vector<std::vector<int32_t>> cmaps;
   PhaseTimer pt;
-  pt.lap("level0 copy", N);
+  pt.lap("level0 dedup", N);
 
   // --- coarsen
   const int64_t coarse_target = std::max<int64_t>(128 * nparts, 4096);
@@ -357,7 +527,7 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
       int64_t cut = 0;
       for (int64_t x = 0; x < gc.n; ++x)
         for (int64_t e = gc.indptr[x]; e < gc.indptr[x + 1]; ++e)
-          if (p[gc.indices[e]] != p[x]) cut += gc.ew[e];
+          if (p[gc.indices[e]] != p[x]) cut += gc.ew(e);
       return cut;
     };
     int64_t best_cut = -1;
@@ -388,6 +558,7 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
                  : std::max<int64_t>(n_refine_passes, 32);
     refine(levels[lvl], nparts, objective, lo_w, cap_w, passes, part);
     pt.lap("project+refine", levels[lvl].n);
+    levels.pop_back();  // the finer level is no longer needed
   }
 
   std::memcpy(out.data_ptr<int32_t>(), part.data(), N * sizeof(int32_t));
