@@ -102,9 +102,33 @@ int64_t heavy_edge_matching(const Graph& g, std::mt19937_64& rng,
     if (best >= 0) {
       match[u] = best;
       match[best] = u;
-    } else {
-      match[u] = u;
     }
+  }
+  // 2-hop pass (METIS-style): heavy-tailed graphs stall the 1-hop pass —
+  // once the hubs are matched, their leaves have no unmatched neighbor
+  // left. Merge pairs of unmatched leaves that SHARE a neighbor: one
+  // deterministic O(E) sweep with a per-node pending-leaf slot.
+  {
+    std::vector<int32_t> pending(n, -1);
+    for (int64_t u = 0; u < n; ++u) {
+      if (match[u] >= 0) continue;
+      bool done = false;
+      for (int64_t e = g.indptr[u]; e < g.indptr[u + 1] && !done; ++e) {
+        const int32_t v = g.indices[e];
+        if (v == u) continue;
+        const int32_t w = pending[v];
+        if (w >= 0 && w != (int32_t)u && match[w] < 0) {
+          match[u] = w;
+          match[w] = (int32_t)u;
+          pending[v] = -1;
+          done = true;
+        } else {
+          pending[v] = (int32_t)u;
+        }
+      }
+    }
+    for (int64_t u = 0; u < n; ++u)
+      if (match[u] < 0) match[u] = (int32_t)u;
   }
   cmap.assign(n, -1);
   int64_t nc = 0;
@@ -289,7 +313,13 @@ int64_t node_vol(const Graph& g, const std::vector<int32_t>& part,
   return g.nw(v) * reps;
 }
 
-// FM-style boundary refinement (single-node moves).
+// FM-style boundary refinement (single-node moves), parallel-evaluated:
+// rounds of read-only gain evaluation over the active-queue snapshot
+// (at::parallel_for against the frozen part[]), then serial in-order
+// application that re-queues any candidate whose neighborhood changed
+// earlier in the same round (staleness guard). Deterministic for a fixed
+// graph+seed regardless of thread count: evaluation is a pure function
+// of the round snapshot and application order is node order.
 //  objective 0: weighted edge-cut gains.
 //  objective 1 ("vol"): exact communication-volume gains when the
 //    nbrcnt table fits (see header comment); approximate otherwise.
@@ -325,120 +355,226 @@ void refine(const Graph& g, int64_t nparts, int64_t objective, int64_t lo_w,
   std::vector<int64_t> vol_scratch;
   if (check) vol_scratch.assign(nparts, 0);
 
-  std::vector<int64_t> cnt(nparts, 0);
-  std::vector<int64_t> addq(nparts, 0);
-  std::vector<int32_t> touched;
-  touched.reserve(64);
-  for (int64_t pass = 0; pass < passes; ++pass) {
-    int64_t moves = 0;
-    for (int64_t x = 0; x < n; ++x) {
-      const int32_t a = part[x];
-      if (psize[a] - g.nw(x) < lo_w) continue;
-      bool boundary = false;
-      for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
-        const int32_t v = g.indices[e];
-        if (v == x) continue;  // self-loops carry no cut/volume
-        const int32_t q = part[v];
-        if (cnt[q] == 0) touched.push_back(q);
-        cnt[q] += g.ew(e);
-        if (q != a) boundary = true;
-      }
-      if (boundary) {
-        int64_t rep_a = 0;  // replicas of x if it stays in a
-        if (objective == 1)
-          for (const int32_t t : touched)
-            if (t != a && cnt[t] > 0) rep_a++;
-        int64_t rem_a = 0;  // nbr volume freed when x leaves a
-        if (exact) {
-          for (const int32_t t : touched) addq[t] = 0;
-          for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
-            const int32_t v = g.indices[e];
-            if (v == x) continue;
-            const int32_t b = part[v];
-            const int32_t* row = nbrcnt.data() + (int64_t)v * nparts;
-            if (b != a && row[a] == 1) rem_a += g.nw(v);
-            for (const int32_t t : touched)
-              if (t != b && row[t] == 0) addq[t] += g.nw(v);
-          }
+  struct Scratch {
+    std::vector<int64_t> cnt, addq;
+    std::vector<int32_t> touched;
+  };
+
+  // gain evaluation for one node against the current (frozen) part[]:
+  // fills best target (== part[x] when no improving move) and its gains
+  auto evaluate = [&](int64_t x, Scratch& s, int32_t& out_best,
+                      int64_t& out_vol, int64_t& out_cut) {
+    const int32_t a = part[x];
+    out_best = a;
+    out_vol = out_cut = 0;
+    if (psize[a] - g.nw(x) < lo_w) return;
+    bool boundary = false;
+    for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+      const int32_t v = g.indices[e];
+      if (v == x) continue;  // self-loops carry no cut/volume
+      const int32_t q = part[v];
+      if (s.cnt[q] == 0) s.touched.push_back(q);
+      s.cnt[q] += g.ew(e);
+      if (q != a) boundary = true;
+    }
+    if (boundary) {
+      int64_t rep_a = 0;  // replicas of x if it stays in a
+      if (objective == 1)
+        for (const int32_t t : s.touched)
+          if (t != a && s.cnt[t] > 0) rep_a++;
+      int64_t rem_a = 0;  // nbr volume freed when x leaves a
+      if (exact) {
+        for (const int32_t t : s.touched) s.addq[t] = 0;
+        for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+          const int32_t v = g.indices[e];
+          if (v == x) continue;
+          const int32_t b = part[v];
+          const int32_t* row = nbrcnt.data() + (int64_t)v * nparts;
+          if (b != a && row[a] == 1) rem_a += g.nw(v);
+          for (const int32_t t : s.touched)
+            if (t != b && row[t] == 0) s.addq[t] += g.nw(v);
         }
-        int32_t best = a;
-        int64_t best_gain = 0, best_vol = 0, best_cut = 0;
-        for (const int32_t q : touched) {
-          if (q == a || psize[q] + g.nw(x) > cap_w) continue;
-          const int64_t cut_gain = cnt[q] - cnt[a];
-          if (objective == 1) {
-            int64_t rep_q = 0;
-            for (const int32_t t : touched)
-              if (t != q && cnt[t] > 0) rep_q++;
-            if (exact) {
-              // exact volume gain: own replicas + neighbor add/remove
-              const int64_t vol_gain =
-                  g.nw(x) * (rep_a - rep_q) + rem_a - addq[q];
-              // lexicographic (volume, cut): volume is the objective,
-              // cut breaks ties; only strictly-improving moves
-              // (vol>0, or vol-neutral with cut>0) are ever accepted
-              if ((vol_gain > best_vol ||
-                   (vol_gain == best_vol && cut_gain > best_cut)) &&
-                  (vol_gain > 0 || (vol_gain == 0 && cut_gain > 0))) {
-                best = q;
-                best_vol = vol_gain;
-                best_cut = cut_gain;
-              }
-              continue;
-            }
-            // approximate: own replica delta only
-            const int64_t gain = cut_gain + rep_a - rep_q;
-            if (gain > best_gain) {
-              best_gain = gain;
-              best = q;
+      }
+      int64_t best_gain = 0;
+      for (const int32_t q : s.touched) {
+        if (q == a || psize[q] + g.nw(x) > cap_w) continue;
+        const int64_t cut_gain = s.cnt[q] - s.cnt[a];
+        if (objective == 1) {
+          int64_t rep_q = 0;
+          for (const int32_t t : s.touched)
+            if (t != q && s.cnt[t] > 0) rep_q++;
+          if (exact) {
+            // exact volume gain: own replicas + neighbor add/remove
+            const int64_t vol_gain =
+                g.nw(x) * (rep_a - rep_q) + rem_a - s.addq[q];
+            // lexicographic (volume, cut): volume is the objective,
+            // cut breaks ties; only strictly-improving moves
+            // (vol>0, or vol-neutral with cut>0) are ever accepted
+            if ((vol_gain > out_vol ||
+                 (vol_gain == out_vol && cut_gain > out_cut)) &&
+                (vol_gain > 0 || (vol_gain == 0 && cut_gain > 0))) {
+              out_best = q;
+              out_vol = vol_gain;
+              out_cut = cut_gain;
             }
             continue;
           }
-          if (cut_gain > best_gain) {
-            best_gain = cut_gain;
-            best = q;
+          // approximate: own replica delta only
+          const int64_t gain = cut_gain + rep_a - rep_q;
+          if (gain > best_gain) {
+            best_gain = gain;
+            out_best = q;
           }
+          continue;
         }
-        if (best != a) {
-          if (check) {
-            // oracle: recompute local volume of {x} u N(x) before/after
-            int64_t before = node_vol(g, part, vol_scratch, x);
-            for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e)
-              if (g.indices[e] != x)
-                before += node_vol(g, part, vol_scratch, g.indices[e]);
-            part[x] = best;
-            int64_t after = node_vol(g, part, vol_scratch, x);
-            for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e)
-              if (g.indices[e] != x)
-                after += node_vol(g, part, vol_scratch, g.indices[e]);
-            part[x] = a;
-            // duplicate neighbors in S would double-count; level-0 is
-            // deduped and coarse graphs are simple, so each neighbor
-            // appears once
-            TORCH_CHECK(before - after == best_vol,
-                        "exact vol gain mismatch: predicted ", best_vol,
-                        " actual ", before - after, " at node ", x);
-          }
-          if (exact) {
-            // maintain neighbor counters for the move a -> best
-            for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
-              const int32_t v = g.indices[e];
-              if (v == x) continue;
-              nbrcnt[(int64_t)v * nparts + a]--;
-              nbrcnt[(int64_t)v * nparts + best]++;
-            }
-          }
-          part[x] = best;
-          psize[a] -= g.nw(x);
-          psize[best] += g.nw(x);
-          moves++;
+        if (cut_gain > best_gain) {
+          best_gain = cut_gain;
+          out_best = q;
         }
       }
-      for (const int32_t q : touched) cnt[q] = 0;
-      touched.clear();
     }
-    if (moves == 0) break;
+    for (const int32_t q : s.touched) s.cnt[q] = 0;
+    s.touched.clear();
+  };
+
+  // active queue, seeded with every boundary node in node order
+  // (parallel detection, chunk-ordered merge keeps it deterministic)
+  std::vector<uint8_t> in_q(n, 0);
+  std::vector<int64_t> work, next;
+  {
+    const int64_t nchunk = 64;
+    const int64_t csz = (n + nchunk - 1) / nchunk;
+    std::vector<std::vector<int64_t>> found(nchunk);
+    at::parallel_for(0, nchunk, 1, [&](int64_t cb, int64_t ce) {
+      for (int64_t c = cb; c < ce; ++c) {
+        for (int64_t x = c * csz; x < std::min(n, (c + 1) * csz); ++x)
+          for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e)
+            if (g.indices[e] != x && part[g.indices[e]] != part[x]) {
+              found[c].push_back(x);
+              break;
+            }
+      }
+    });
+    for (auto& f : found)
+      for (int64_t x : f) {
+        work.push_back(x);
+        in_q[x] = 1;
+      }
   }
+  std::vector<int32_t> bestv;
+  // evaluation budget guards the approximate mode (its gain is heuristic,
+  // so the true objective need not strictly decrease); exact mode
+  // terminates on its own (lexicographic (vol,cut) strictly decreases)
+  int64_t pops_left = std::max<int64_t>(passes, 4) * std::max<int64_t>(n, 1);
+  // `passes` doubles as the round cap: each round is one parallel sweep
+  // of the active queue (the FM-pass analogue); refinement gains decay
+  // fast with rounds, so the cap trades a few % of volume for bounded
+  // O(rounds * boundary) work at 100M+ nodes
+  const int64_t cap_env = env_int("PIPEGCN_PART_ROUNDS", 0);
+  const int64_t rounds_cap =
+      cap_env > 0 ? cap_env
+                  : (passes >= 999 ? (int64_t)1 << 40
+                                   : std::max<int64_t>(passes, 2));
+  int64_t n_rounds = 0, n_pre = 0, n_live = 0, n_moves = 0;
+  double t_par = 0, t_ser = 0;
+  using clk = std::chrono::steady_clock;
+
+  while (!work.empty() && pops_left > 0 && n_rounds < rounds_cap) {
+    ++n_rounds;
+    const int64_t R = std::min<int64_t>((int64_t)work.size(), pops_left);
+    pops_left -= R;
+    bestv.assign(R, 0);
+    // parallel PREFILTER: gains against the round-start snapshot select
+    // candidates; the serial pass below re-evaluates each candidate on
+    // the LIVE state before applying, so a stale prefilter result can
+    // cost a wasted re-evaluation but never a wrong move. Small rounds
+    // run inline (parallel_for launch overhead dwarfs tiny batches).
+    auto eval_range = [&](int64_t b, int64_t en) {
+      Scratch s;
+      s.cnt.assign(nparts, 0);
+      s.addq.assign(nparts, 0);
+      s.touched.reserve(64);
+      int64_t vol_g, cut_g;
+      for (int64_t i = b; i < en; ++i)
+        evaluate(work[i], s, bestv[i], vol_g, cut_g);
+    };
+    auto tp0 = clk::now();
+    if (R < 8192)
+      eval_range(0, R);
+    else
+      at::parallel_for(0, R, 4096, eval_range);
+    auto tp1 = clk::now();
+    t_par += std::chrono::duration<double>(tp1 - tp0).count();
+    next.clear();
+    auto push_next = [&](int64_t v) {
+      if (!in_q[v]) {
+        in_q[v] = 1;
+        next.push_back(v);
+      }
+    };
+    Scratch s;
+    s.cnt.assign(nparts, 0);
+    s.addq.assign(nparts, 0);
+    s.touched.reserve(64);
+    for (int64_t i = 0; i < R; ++i) {
+      const int64_t x = work[i];
+      in_q[x] = 0;
+      if (bestv[i] == part[x]) continue;  // prefilter: no move candidate
+      // live re-evaluation (neighbors/psize may have changed since the
+      // snapshot; exact gains also depend on 2-hop state via the
+      // neighbors' counter rows, so only a live gain is trustworthy)
+      const int32_t a = part[x];
+      int32_t best;
+      int64_t vol_gain, cut_gain;
+      ++n_live;
+      evaluate(x, s, best, vol_gain, cut_gain);
+      if (best == a) continue;
+      ++n_moves;
+      if (check) {
+        // oracle: recompute local volume of {x} u N(x) before/after
+        int64_t before = node_vol(g, part, vol_scratch, x);
+        for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e)
+          if (g.indices[e] != x)
+            before += node_vol(g, part, vol_scratch, g.indices[e]);
+        part[x] = best;
+        int64_t after = node_vol(g, part, vol_scratch, x);
+        for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e)
+          if (g.indices[e] != x)
+            after += node_vol(g, part, vol_scratch, g.indices[e]);
+        part[x] = a;
+        TORCH_CHECK(before - after == vol_gain,
+                    "exact vol gain mismatch: predicted ", vol_gain,
+                    " actual ", before - after, " at node ", x);
+      }
+      if (exact) {
+        // maintain neighbor counters for the move a -> best
+        for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+          const int32_t v = g.indices[e];
+          if (v == x) continue;
+          nbrcnt[(int64_t)v * nparts + a]--;
+          nbrcnt[(int64_t)v * nparts + best]++;
+        }
+      }
+      part[x] = best;
+      psize[a] -= g.nw(x);
+      psize[best] += g.nw(x);
+      // the move changed every neighbor's gain landscape (and x's own)
+      for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+        const int32_t v = g.indices[e];
+        if (v != x) push_next(v);
+      }
+      push_next(x);
+    }
+    n_pre += R;
+    t_ser += std::chrono::duration<double>(clk::now() - tp1).count();
+    std::swap(work, next);
+  }
+  if (std::getenv("PIPEGCN_PART_VERBOSE"))
+    fprintf(stderr,
+            "[partition]   refine n=%lld rounds=%lld pre=%lld live=%lld "
+            "moves=%lld par=%.2fs ser=%.2fs\n",
+            (long long)n, (long long)n_rounds, (long long)n_pre,
+            (long long)n_live, (long long)n_moves, t_par, t_ser);
 }
 
 }  // namespace
@@ -501,12 +637,26 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
   pt.lap("level0 dedup", N);
 
   // --- coarsen
-  const int64_t coarse_target = std::max<int64_t>(128 * nparts, 4096);
+  // stop coarsening early: community graphs densify as they coarsen
+  // (E shrinks much slower than n), which makes both coarsen() and the
+  // per-eval O(deg) refinement cost blow up on the small-n levels, and
+  // every extra level adds projection error the finer levels must repair
+  const int64_t coarse_target = std::max<int64_t>(
+      env_int("PIPEGCN_PART_COARSE_N", 65536), 128 * nparts);
   while (levels.back().n > coarse_target && levels.size() < 24) {
     std::vector<int32_t> cmap;
     const int64_t nc = heavy_edge_matching(levels.back(), rng, cmap);
     if (nc > levels.back().n * 95 / 100) break;  // matching stalled
+    // densification guard: stop when halving n no longer sheds edges
+    const int64_t e_prev = (int64_t)levels.back().indices.size();
     levels.push_back(coarsen(levels.back(), cmap, nc));
+    if ((int64_t)levels.back().indices.size() > e_prev * 85 / 100 &&
+        nc < 4 * coarse_target) {
+      // keep the level (already built) but stop here
+      cmaps.push_back(std::move(cmap));
+      pt.lap("match+coarsen", nc);
+      break;
+    }
     cmaps.push_back(std::move(cmap));
     pt.lap("match+coarsen", nc);
   }
@@ -530,18 +680,25 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
           if (p[gc.indices[e]] != p[x]) cut += gc.ew(e);
       return cut;
     };
+    // restart count scales down with coarsest size (matching can stall
+    // well above the target on heavy-tailed graphs; 8 full restarts on a
+    // 100k+-node coarsest level would dominate the whole partition)
+    const int64_t trials =
+        std::max<int64_t>(2, std::min<int64_t>(8, 8 * 8192 / gc.n));
     int64_t best_cut = -1;
-    for (int trial = 0; trial < 8; ++trial) {
+    for (int64_t trial = 0; trial < trials; ++trial) {
       std::vector<int32_t> cand;
       initial_partition(gc, nparts, cap_w, rng, cand);
-      refine(gc, nparts, objective, lo_w, cap_w,
-             std::max<int64_t>(n_refine_passes, 64), cand);
+      refine(gc, nparts, objective, lo_w, cap_w, 10, cand);
       const int64_t c = cut_of(cand);
       if (best_cut < 0 || c < best_cut) {
         best_cut = c;
         part = std::move(cand);
       }
     }
+    // polish only the winner with the full budget
+    refine(gc, nparts, objective, lo_w, cap_w,
+           std::max<int64_t>(n_refine_passes, 32), part);
   }
 
   pt.lap("coarsest solve", levels.back().n);
@@ -554,8 +711,8 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
     for (int64_t u = 0; u < levels[lvl].n; ++u) fine[u] = part[cmap[u]];
     part = std::move(fine);
     const int64_t passes =
-        lvl == 0 ? std::max<int64_t>(n_refine_passes / 2, 2)
-                 : std::max<int64_t>(n_refine_passes, 32);
+        lvl == 0 ? std::max<int64_t>(n_refine_passes / 2, 4)
+                 : std::max<int64_t>(n_refine_passes, 16);
     refine(levels[lvl], nparts, objective, lo_w, cap_w, passes, part);
     pt.lap("project+refine", levels[lvl].n);
     levels.pop_back();  // the finer level is no longer needed

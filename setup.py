@@ -26,9 +26,13 @@ ext = CUDAExtension(
         "pipegcn_amd/csrc/hip/elementwise.hip",
     ],
     extra_compile_args={
-        "cxx": ["-O3", "-std=c++17"],
+        # -fopenmp: at::parallel_for is a header template — its OpenMP
+        # pragmas expand in THIS translation unit and silently serialize
+        # without the flag (measured: 8-thread partitioner phases ran 1x)
+        "cxx": ["-O3", "-std=c++17", "-fopenmp"],
         "nvcc": ["-O3", "-std=c++17"],
     },
+    extra_link_args=["-fopenmp"],
 )
 
 setup(
