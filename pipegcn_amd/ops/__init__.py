@@ -121,10 +121,21 @@ class _SageDualLinear(torch.autograd.Function):
     def backward(ctx, g):
         x1, x2, w1, w2 = ctx.saved_tensors
         g = g.contiguous()
-        gx1 = g @ w1
-        gx2 = g @ w2
-        gw1 = g.t() @ x1
-        gw2 = g.t() @ x2
+        if g.is_cuda and g.dtype == torch.float32:
+            # dgrad pair as ONE GEMM against [w1 ‖ w2] — rocBLAS reads g
+            # once instead of twice (the concat is 2x616 KB, free)
+            gx = g @ torch.cat((w1, w2), dim=1)
+            K = w1.size(1)
+            gx1, gx2 = gx[:, :K], gx[:, K:]
+            # wgrad pair fused in one MFMA split-M kernel (g streamed once
+            # for both products; deterministic workspace reduce)
+            gw1, gw2 = native().dual_wgrad(g, x1.contiguous(),
+                                           x2.contiguous())
+        else:
+            gx1 = g @ w1
+            gx2 = g @ w2
+            gw1 = g.t() @ x1
+            gw2 = g.t() @ x2
         if not ctx.has_bias:
             return gx1, gx2, gw1, gw2, None, None
         gb = native().colsum(g)  # two-phase column sum (fastest measured)
@@ -222,7 +233,11 @@ class _LinearColsum(torch.autograd.Function):
         x, w = ctx.saved_tensors
         g = g.contiguous()
         gb = native().colsum(g).to(w.dtype)
-        return g @ w, g.t() @ x, gb
+        if g.dtype == torch.float32:
+            (gw,) = native().dual_wgrad(g, x.contiguous(), torch.Tensor())
+        else:
+            gw = g.t() @ x
+        return g @ w, gw, gb
 
 
 def linear(x, lin):
