@@ -81,10 +81,31 @@ __global__ void dropout_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     }
     mask[t] = m;
     if (i0 + 8 <= n) {
+      if constexpr (sizeof(T) == 4) {
+        // fp32: two float4 transactions instead of 8 scalar dwords
+        // (i0 = 8t so the 16B alignment always holds)
+        const float4 a = *reinterpret_cast<const float4*>(
+            reinterpret_cast<const float*>(x) + i0);
+        const float4 b = *reinterpret_cast<const float4*>(
+            reinterpret_cast<const float*>(x) + i0 + 4);
+        float4 oa, ob;
+        oa.x = (m & 1u) ? a.x * scale : 0.f;
+        oa.y = (m & 2u) ? a.y * scale : 0.f;
+        oa.z = (m & 4u) ? a.z * scale : 0.f;
+        oa.w = (m & 8u) ? a.w * scale : 0.f;
+        ob.x = (m & 16u) ? b.x * scale : 0.f;
+        ob.y = (m & 32u) ? b.y * scale : 0.f;
+        ob.z = (m & 64u) ? b.z * scale : 0.f;
+        ob.w = (m & 128u) ? b.w * scale : 0.f;
+        *reinterpret_cast<float4*>(reinterpret_cast<float*>(y) + i0) = oa;
+        *reinterpret_cast<float4*>(reinterpret_cast<float*>(y) + i0 + 4) =
+            ob;
+      } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float v = (m >> j) & 1 ? to_f32(x[i0 + j]) * scale : 0.f;
-        from_f32(v, &y[i0 + j]);
+        for (int j = 0; j < 8; ++j) {
+          const float v = (m >> j) & 1 ? to_f32(x[i0 + j]) * scale : 0.f;
+          from_f32(v, &y[i0 + j]);
+        }
       }
     } else {
       for (int j = 0; i0 + j < n; ++j) {
@@ -107,10 +128,29 @@ __global__ void dropout_bwd_kernel(const T* __restrict__ dy,
     const int64_t i0 = t * 8;
     const uint8_t m = mask[t];
     if (i0 + 8 <= n) {
+      if constexpr (sizeof(T) == 4) {
+        const float4 a = *reinterpret_cast<const float4*>(
+            reinterpret_cast<const float*>(dy) + i0);
+        const float4 b = *reinterpret_cast<const float4*>(
+            reinterpret_cast<const float*>(dy) + i0 + 4);
+        float4 oa, ob;
+        oa.x = (m & 1u) ? a.x * scale : 0.f;
+        oa.y = (m & 2u) ? a.y * scale : 0.f;
+        oa.z = (m & 4u) ? a.z * scale : 0.f;
+        oa.w = (m & 8u) ? a.w * scale : 0.f;
+        ob.x = (m & 16u) ? b.x * scale : 0.f;
+        ob.y = (m & 32u) ? b.y * scale : 0.f;
+        ob.z = (m & 64u) ? b.z * scale : 0.f;
+        ob.w = (m & 128u) ? b.w * scale : 0.f;
+        *reinterpret_cast<float4*>(reinterpret_cast<float*>(dx) + i0) = oa;
+        *reinterpret_cast<float4*>(reinterpret_cast<float*>(dx) + i0 + 4) =
+            ob;
+      } else {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float v = (m >> j) & 1 ? to_f32(dy[i0 + j]) * scale : 0.f;
-        from_f32(v, &dx[i0 + j]);
+        for (int j = 0; j < 8; ++j) {
+          const float v = (m >> j) & 1 ? to_f32(dy[i0 + j]) * scale : 0.f;
+          from_f32(v, &dx[i0 + j]);
+        }
       }
     } else {
       for (int j = 0; i0 + j < n; ++j) {

@@ -636,29 +636,65 @@ torch::Tensor partition_graph_cpu(torch::Tensor indptr, torch::Tensor indices,
   PhaseTimer pt;
   pt.lap("level0 dedup", N);
 
-  // --- coarsen
+  // --- coarsen (memory-bounded level retention)
   // stop coarsening early: community graphs densify as they coarsen
   // (E shrinks much slower than n), which makes both coarsen() and the
   // per-eval O(deg) refinement cost blow up on the small-n levels, and
-  // every extra level adds projection error the finer levels must repair
+  // every extra level adds projection error the finer levels must repair.
+  // Dense MIDDLE levels of a 100M+-node graph can hold >1B edges EACH;
+  // retaining every one for uncoarsening refinement OOMs long before the
+  // coarsest solve (measured: 65 GB RSS at papers100M scale). Levels
+  // whose edge count exceeds PIPEGCN_PART_KEEP_EDGES (default 400M) are
+  // used for matching and then DISCARDED: their projection maps compose,
+  // and refinement happens only at the retained levels (level 0 always).
   const int64_t coarse_target = std::max<int64_t>(
       env_int("PIPEGCN_PART_COARSE_N", 65536), 128 * nparts);
-  while (levels.back().n > coarse_target && levels.size() < 24) {
-    std::vector<int32_t> cmap;
-    const int64_t nc = heavy_edge_matching(levels.back(), rng, cmap);
-    if (nc > levels.back().n * 95 / 100) break;  // matching stalled
-    // densification guard: stop when halving n no longer sheds edges
-    const int64_t e_prev = (int64_t)levels.back().indices.size();
-    levels.push_back(coarsen(levels.back(), cmap, nc));
-    if ((int64_t)levels.back().indices.size() > e_prev * 85 / 100 &&
-        nc < 4 * coarse_target) {
-      // keep the level (already built) but stop here
-      cmaps.push_back(std::move(cmap));
-      pt.lap("match+coarsen", nc);
-      break;
+  const int64_t keep_edges =
+      env_int("PIPEGCN_PART_KEEP_EDGES", 400000000);
+  {
+    std::vector<int32_t> pending;  // levels.back() ids -> front ids
+    Graph front_store;             // the front when it is not retained
+    int64_t n_levels_total = 1;
+    while (true) {
+      Graph& front = pending.empty() ? levels.back() : front_store;
+      if (front.n <= coarse_target || n_levels_total >= 24) break;
+      std::vector<int32_t> cmap;
+      const int64_t nc = heavy_edge_matching(front, rng, cmap);
+      if (nc > front.n * 95 / 100) break;  // matching stalled
+      const int64_t e_prev = (int64_t)front.indices.size();
+      Graph next = coarsen(front, cmap, nc);
+      n_levels_total++;
+      // densification guard: stop when halving n no longer sheds edges
+      const bool densified =
+          (int64_t)next.indices.size() > e_prev * 85 / 100 &&
+          nc < 4 * coarse_target;
+      if (pending.empty()) {
+        pending = std::move(cmap);
+      } else {
+        at::parallel_for(0, (int64_t)pending.size(), 1 << 16,
+                         [&](int64_t b, int64_t en) {
+                           for (int64_t i = b; i < en; ++i)
+                             pending[i] = cmap[pending[i]];
+                         });
+      }
+      const bool retain = (int64_t)next.indices.size() <= keep_edges ||
+                          densified;
+      if (retain) {
+        levels.push_back(std::move(next));
+        cmaps.push_back(std::move(pending));
+        pending.clear();
+        front_store = Graph();  // free any unretained front
+      } else {
+        front_store = std::move(next);
+      }
+      pt.lap(retain ? "match+coarsen" : "match+coarsen drop", nc);
+      if (densified) break;
     }
-    cmaps.push_back(std::move(cmap));
-    pt.lap("match+coarsen", nc);
+    // the coarsening front is the coarsest-solve target: retain it
+    if (!pending.empty()) {
+      levels.push_back(std::move(front_store));
+      cmaps.push_back(std::move(pending));
+    }
   }
 
   // --- initial partition on the coarsest level
