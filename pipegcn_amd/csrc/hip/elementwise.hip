@@ -176,47 +176,69 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x,
                               T* __restrict__ y, T* __restrict__ xhat,
                               float* __restrict__ rstd_out, float eps,
                               int64_t N, int F) {
+  // TWO rows per wave: the 12 cross-lane reduction steps per row are
+  // latency-bound, and two independent rows interleave them (measured
+  // win over one-row-per-wave at [233k,256])
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int64_t row = static_cast<int64_t>(blockIdx.x) * 4 + wave;
-  if (row >= N) return;
-  const T* xr = x + row * F;
+  const int64_t row0 = (static_cast<int64_t>(blockIdx.x) * 4 + wave) * 2;
+  if (row0 >= N) return;
+  const int nrows = row0 + 1 < N ? 2 : 1;
 
-  float v[LN_MAXCH][LN_VEC];
-  float s = 0.f, s2 = 0.f;
+  float v[2][LN_MAXCH][LN_VEC];
+  float s[2] = {0.f, 0.f}, s2[2] = {0.f, 0.f};
   const int nch = (F + 64 * LN_VEC - 1) / (64 * LN_VEC);
 #pragma unroll
-  for (int c = 0; c < LN_MAXCH; ++c) {
-    if (c >= nch) break;
+  for (int r = 0; r < 2; ++r) {
+    if (r >= nrows) break;
+    const T* xr = x + (row0 + r) * F;
 #pragma unroll
-    for (int k = 0; k < LN_VEC; ++k) {
-      const int f = (c * 64 + lane) * LN_VEC + k;
-      const float u = f < F ? to_f32(xr[f]) : 0.f;
-      v[c][k] = u;
-      s += u;
-      s2 += u * u;
+    for (int c = 0; c < LN_MAXCH; ++c) {
+      if (c >= nch) break;
+#pragma unroll
+      for (int k = 0; k < LN_VEC; ++k) {
+        const int f = (c * 64 + lane) * LN_VEC + k;
+        const float u = f < F ? to_f32(xr[f]) : 0.f;
+        v[r][c][k] = u;
+        s[r] += u;
+        s2[r] += u * u;
+      }
     }
   }
-  s = wave_sum(s);
-  s2 = wave_sum(s2);
-  const float mean = s / F;
-  const float rstd = rsqrtf(fmaxf(s2 / F - mean * mean, 0.f) + eps);
-  if (lane == 0) rstd_out[row] = rstd;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s[0] += __shfl_xor(s[0], off, 64);
+    s2[0] += __shfl_xor(s2[0], off, 64);
+    s[1] += __shfl_xor(s[1], off, 64);
+    s2[1] += __shfl_xor(s2[1], off, 64);
+  }
+  float mean[2], rstd[2];
+#pragma unroll
+  for (int r = 0; r < 2; ++r) {
+    if (r >= nrows) break;
+    mean[r] = s[r] / F;
+    rstd[r] = rsqrtf(fmaxf(s2[r] / F - mean[r] * mean[r], 0.f) + eps);
+    if (lane == 0) rstd_out[row0 + r] = rstd[r];
+  }
 
-  T* yr = y + row * F;
-  T* hr = xhat + row * F;
 #pragma unroll
-  for (int c = 0; c < LN_MAXCH; ++c) {
-    if (c >= nch) break;
+  for (int r = 0; r < 2; ++r) {
+    if (r >= nrows) break;
+    T* yr = y + (row0 + r) * F;
+    T* hr = xhat + (row0 + r) * F;
 #pragma unroll
-    for (int k = 0; k < LN_VEC; ++k) {
-      const int f = (c * 64 + lane) * LN_VEC + k;
-      if (f >= F) break;
-      const float h = (v[c][k] - mean) * rstd;
-      from_f32(h, &hr[f]);
-      float o = w[f] * h + b[f];
-      if (RELU) o = fmaxf(o, 0.f);
-      from_f32(o, &yr[f]);
+    for (int c = 0; c < LN_MAXCH; ++c) {
+      if (c >= nch) break;
+#pragma unroll
+      for (int k = 0; k < LN_VEC; ++k) {
+        const int f = (c * 64 + lane) * LN_VEC + k;
+        if (f >= F) break;
+        const float h = (v[r][c][k] - mean[r]) * rstd[r];
+        from_f32(h, &hr[f]);
+        float o = w[f] * h + b[f];
+        if (RELU) o = fmaxf(o, 0.f);
+        from_f32(o, &yr[f]);
+      }
     }
   }
 }
@@ -383,7 +405,7 @@ void layer_norm_relu_fwd_hip(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(w.is_contiguous() && b.is_contiguous());
   TORCH_CHECK(rstd.scalar_type() == torch::kFloat && rstd.numel() == N);
   if (N == 0) return;
-  const int64_t blocks = (N + 3) / 4;
+  const int64_t blocks = (N + 7) / 8;  // 4 waves x 2 rows per block
   const bool bf16 = x.scalar_type() == torch::kBFloat16;
   TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat);
   auto launch = [&](auto tptr, auto kernel) {

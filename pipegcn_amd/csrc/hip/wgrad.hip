@@ -14,9 +14,9 @@
 //
 // Uses v_mfma_f32_32x32x2_f32 (f32 in / f32 accumulate — an exact fmaf
 // chain, no TF32; cdna_hip_programming.md §3). Block: 256 threads,
-// BNxBK output tile (128x128 when the output is big enough — halves
-// both the g and the x tile re-reads vs 64x64 — measured 66->96 TF at
-// the [233k]x[256,602] reddit shape), 4 waves in a 2x2 quadrant layout,
+// BNxBK output tile (64x64 default; see the host-side note on why the
+// lower-traffic 128x128 variant measured SLOWER), 4 waves in a 2x2
+// quadrant layout,
 // FN x FK 32x32 fragments per wave per product; M staged through LDS 32
 // rows at a time (+1-dword row padding against bank conflicts).
 // x2 == nullptr computes gw1 only (tail linears, GCN).
