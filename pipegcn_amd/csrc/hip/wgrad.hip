@@ -30,11 +30,6 @@
 
 namespace {
 
-inline bool env_flag_128() {
-  const char* s = std::getenv("PIPEGCN_WGRAD_128");
-  return s && s[0] == '1';
-}
-
 using f32x16 = __attribute__((__vector_size__(16 * sizeof(float)))) float;
 
 template <bool DUAL, int BN, int BK>
@@ -180,9 +175,11 @@ std::vector<torch::Tensor> dual_wgrad_hip(torch::Tensor g, torch::Tensor x1,
   // and LOSE (66->57 TF at [233k]x[256,602], 111->100 TF at K=256) —
   // the kernel is latency-, not bandwidth-, bound at these shapes.
   // Template retained for the record; env knob for re-measurement.
-  const bool big = env_flag_128();
-  const int BN = big ? 128 : 64;
-  const int BK = big ? 128 : 64;
+  // PIPEGCN_WGRAD_CFG: 0 (default) 64x64, 1 -> 128x128, 2 -> 64x128
+  const char* cfg_s = std::getenv("PIPEGCN_WGRAD_CFG");
+  const int cfg = cfg_s ? std::atoi(cfg_s) : 0;
+  const int BN = cfg == 1 ? 128 : 64;
+  const int BK = cfg >= 1 ? 128 : 64;
   const int64_t tiles = ((N + BN - 1) / BN) * ((K + BK - 1) / BK);
   // enough blocks to fill 256 CUs several times over, M chunks 32-aligned
   int64_t S = std::min<int64_t>((1024 + tiles - 1) / tiles, 64);
@@ -205,13 +202,17 @@ std::vector<torch::Tensor> dual_wgrad_hip(torch::Tensor g, torch::Tensor x1,
                      grid, dim3(256), 0, stream, gp, x1p, x2p, w1p, w2p, \
                      M, N, K, chunk)
   if (dual) {
-    if (big)
+    if (cfg == 1)
       LAUNCH_WGRAD(true, 128, 128);
+    else if (cfg == 2)
+      LAUNCH_WGRAD(true, 64, 128);
     else
       LAUNCH_WGRAD(true, 64, 64);
   } else {
-    if (big)
+    if (cfg == 1)
       LAUNCH_WGRAD(false, 128, 128);
+    else if (cfg == 2)
+      LAUNCH_WGRAD(false, 64, 128);
     else
       LAUNCH_WGRAD(false, 64, 64);
   }
