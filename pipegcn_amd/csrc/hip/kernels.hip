@@ -75,7 +75,8 @@ inline hipStream_t current_stream() {
 // fastest (CHUNK_OUTER=false) every panel streams concurrently and the
 // combined working set thrashes L3.  HAS_SRC_SCALE fuses a per-source-row
 // scale (the transpose/backward SpMM's D^{-1} pre-scale) into the gather.
-template <typename T, int VEC, bool CHUNK_OUTER, bool HAS_SRC_SCALE>
+template <typename T, int VEC, bool CHUNK_OUTER, bool HAS_SRC_SCALE,
+          bool U8 = false>
 __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
                                 const int32_t* __restrict__ indices,
                                 const T* __restrict__ feat,
@@ -126,6 +127,35 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
 #pragma unroll
         for (int k = 0; k < VEC; ++k)
           acc[k] += s * to_f32(feat[u * F + f0 + k]);
+      }
+      if (U8) {
+        // 8 outstanding gathers per wave instead of 4: PMC showed the
+        // 4-edge loop reaches only ~3.8 TB/s HBM (6.6% L2 hit) — the
+        // gather stream is memory-LATENCY-bound, and doubling the
+        // in-flight requests per wave is the lever occupancy alone
+        // could not supply (see profiles/README.md round 2)
+        for (; e + 8 <= e_end; e += 8) {
+          const int32x4 ua = __builtin_nontemporal_load(
+              reinterpret_cast<const int32x4*>(indices + e));
+          const int32x4 ub = __builtin_nontemporal_load(
+              reinterpret_cast<const int32x4*>(indices + e + 4));
+          const int64_t u[8] = {ua[0], ua[1], ua[2], ua[3],
+                                ub[0], ub[1], ub[2], ub[3]};
+          float sc[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            sc[j] = HAS_SRC_SCALE ? src_scale[u[j]] : 1.f;
+          float v[8][VEC];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+#pragma unroll
+            for (int k = 0; k < VEC; ++k)
+              v[j][k] = to_f32(feat[u[j] * F + f0 + k]);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+#pragma unroll
+            for (int k = 0; k < VEC; ++k) acc[k] += sc[j] * v[j][k];
+        }
       }
       for (; e + 4 <= e_end; e += 4) {
         const int32x4 uu = __builtin_nontemporal_load(
@@ -315,16 +345,26 @@ void launch_spmm(const int64_t* indptr, const int32_t* indices,
                        indices, feat, dst_scale, src_scale, row_order, out,
                        num_rows, F, nchunks);
   };
+  // PIPEGCN_SPMM_U4=1 drops back to the 4-edge inner loop (A/B knob)
+  static const bool u8 = [] {
+    const char* s = std::getenv("PIPEGCN_SPMM_U4");
+    return !(s && s[0] == '1');
+  }();
   if (chunk_outer) {
     if (src_scale)
-      launch(HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, true, true>));
+      launch(u8 ? HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, true, true, true>)
+                : HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, true, true>));
     else
-      launch(HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, true, false>));
+      launch(u8 ? HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, true, false, true>)
+                : HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, true, false>));
   } else {
     if (src_scale)
-      launch(HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, false, true>));
+      launch(u8 ? HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, false, true, true>)
+                : HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, false, true>));
     else
-      launch(HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, false, false>));
+      launch(u8
+                 ? HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, false, false, true>)
+                 : HIP_KERNEL_NAME(spmm_csr_kernel<T, VEC, false, false>));
   }
   HIP_CHECK(hipGetLastError());
 }
