@@ -129,11 +129,8 @@ __global__ void spmm_csr_kernel(const int64_t* __restrict__ indptr,
           acc[k] += s * to_f32(feat[u * F + f0 + k]);
       }
       if (U8) {
-        // 8 outstanding gathers per wave instead of 4: PMC showed the
-        // 4-edge loop reaches only ~3.8 TB/s HBM (6.6% L2 hit) — the
-        // gather stream is memory-LATENCY-bound, and doubling the
-        // in-flight requests per wave is the lever occupancy alone
-        // could not supply (see profiles/README.md round 2)
+        // 8 outstanding gathers per wave instead of 4 (recorded
+        // experiment, PIPEGCN_SPMM_U8 — see host-side note: neutral)
         for (; e + 8 <= e_end; e += 8) {
           const int32x4 ua = __builtin_nontemporal_load(
               reinterpret_cast<const int32x4*>(indices + e));
@@ -345,10 +342,15 @@ void launch_spmm(const int64_t* indptr, const int32_t* indices,
                        indices, feat, dst_scale, src_scale, row_order, out,
                        num_rows, F, nchunks);
   };
-  // PIPEGCN_SPMM_U4=1 drops back to the 4-edge inner loop (A/B knob)
+  // PIPEGCN_SPMM_U8=1 switches to the 8-edge inner loop — measured
+  // NEUTRAL-to-slightly-negative (F=602 fwd 41.2 -> 42.1 ms, F=256
+  // 15.5 -> 15.4): the compiler already software-pipelines the 4-edge
+  // loop's gathers, so per-wave MLP was not the limiter; the 3.8 TB/s
+  // actual-HBM plateau (PMC, profiles/README.md) is DRAM efficiency on
+  // random 512 B row-slice reads, not issue or in-flight depth.
   static const bool u8 = [] {
-    const char* s = std::getenv("PIPEGCN_SPMM_U4");
-    return !(s && s[0] == '1');
+    const char* s = std::getenv("PIPEGCN_SPMM_U8");
+    return s && s[0] == '1';
   }();
   if (chunk_outer) {
     if (src_scale)
