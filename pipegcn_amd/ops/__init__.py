@@ -48,9 +48,13 @@ class _SpmmMean(torch.autograd.Function):
     def backward(ctx, grad_out: torch.Tensor):
         (inv_deg,) = ctx.saved_tensors
         g = ctx.graph
-        # transpose SpMM with the D^{-1} pre-scale fused as a source scale
-        grad_feat = spmm(g.csc, grad_out.contiguous(), None,
-                         src_scale=inv_deg)
+        # D^{-1} applied as ONE streaming row-multiply on g (~0.06 ms at
+        # [233k,256]) instead of a per-EDGE src_scale gather inside the
+        # transpose SpMM (measured ~1 ms/call: fwd-no-scale 15.4 ms vs
+        # bwd-with-src-scale 16.3 ms on the same CSC shape)
+        grad_feat = spmm(g.csc,
+                         grad_out * inv_deg.unsqueeze(1).to(grad_out.dtype),
+                         None)
         return None, grad_feat, None
 
 
