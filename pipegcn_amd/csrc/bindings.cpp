@@ -145,6 +145,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum", &colsum, "out[n] = sum_m x[m,n]");
   m.def("sage_dual_gemm", &sage_dual_gemm,
         "out = x1 @ w1^T + x2 @ w2^T + bias (MFMA fp32, fused)");
+  m.def("dual_dgrad", &dual_dgrad_hip,
+        "gx1 = g w1, gx2 = g w2 fused MFMA dgrad");
   m.def("dual_wgrad", &dual_wgrad_hip,
         "gw1 = g^T x1 [, gw2 = g^T x2] fused MFMA split-M wgrad");
   m.def("dropout_fwd", &dropout_fwd,

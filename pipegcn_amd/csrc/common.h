@@ -81,6 +81,11 @@ void layer_norm_relu_fwd_hip(torch::Tensor x, torch::Tensor w,
                              torch::Tensor y, torch::Tensor xhat,
                              torch::Tensor rstd);
 // dw_part/db_part: fp32 [nwaves, F] per-wave partials (column-sum on host).
+// fused dual data gradient: {g @ w1, g @ w2} via MFMA
+// (csrc/hip/dual_dgrad.hip)
+std::vector<torch::Tensor> dual_dgrad_hip(torch::Tensor g, torch::Tensor w1,
+                                          torch::Tensor w2);
+
 // fused dual weight gradient: {g^T x1 [, g^T x2]} via MFMA split-M
 // (csrc/hip/wgrad.hip); pass an undefined x2 for the single-wgrad case
 std::vector<torch::Tensor> dual_wgrad_hip(torch::Tensor g, torch::Tensor x1,

@@ -98,9 +98,11 @@ def ema_update(avg: torch.Tensor, x: torch.Tensor, momentum: float) -> None:
 class _SageDualLinear(torch.autograd.Function):
     """out = x1 @ w1^T + x2 @ w2^T + (b1 + b2).
 
-    Forward runs the hand-written gfx950 MFMA fused dual-GEMM
-    (csrc/hip/dual_gemm.hip); backward uses rocBLAS GEMMs (dgrad/wgrad are
-    plain library shapes).
+    The whole dense path is hand-written gfx950 MFMA: forward fused
+    dual-GEMM (csrc/hip/dual_gemm.hip), backward fused dual-dgrad
+    (dual_dgrad.hip) + fused split-M dual-wgrad (wgrad.hip) + native
+    colsum bias grad; rocBLAS only on thin-N (output-layer) shapes where
+    the 32-wide MFMA tiles would idle.
     """
 
     @staticmethod
@@ -126,11 +128,17 @@ class _SageDualLinear(torch.autograd.Function):
         x1, x2, w1, w2 = ctx.saved_tensors
         g = g.contiguous()
         if g.is_cuda and g.dtype == torch.float32:
-            # dgrad pair as ONE GEMM against [w1 ‖ w2] — rocBLAS reads g
-            # once instead of twice (the concat is 2x616 KB, free)
-            gx = g @ torch.cat((w1, w2), dim=1)
-            K = w1.size(1)
-            gx1, gx2 = gx[:, :K], gx[:, K:]
+            if g.size(1) >= 64:
+                # dgrad pair fused in one MFMA kernel (g tile staged once
+                # for both weight contractions, contiguous outputs)
+                gx1, gx2 = native().dual_dgrad(g, w1.contiguous(),
+                                               w2.contiguous())
+            else:
+                # thin-N (e.g. the 41-class output layer): rocBLAS wins —
+                # ONE GEMM against [w1 ‖ w2] so g is still read once
+                gx = g @ torch.cat((w1, w2), dim=1)
+                K = w1.size(1)
+                gx1, gx2 = gx[:, :K], gx[:, K:]
             # wgrad pair fused in one MFMA split-M kernel (g streamed once
             # for both products; deterministic workspace reduce)
             gw1, gw2 = native().dual_wgrad(g, x1.contiguous(),

@@ -25,6 +25,7 @@ ext = CUDAExtension(
         "pipegcn_amd/csrc/hip/dual_gemm.hip",
         "pipegcn_amd/csrc/hip/elementwise.hip",
         "pipegcn_amd/csrc/hip/wgrad.hip",
+        "pipegcn_amd/csrc/hip/dual_dgrad.hip",
     ],
     extra_compile_args={
         # -fopenmp: at::parallel_for is a header template — its OpenMP

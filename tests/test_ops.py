@@ -360,3 +360,21 @@ def test_dual_linear_backward_uses_fused_wgrad():
                  (lin1.bias.grad, l1r.bias.grad)):
         assert torch.allclose(a, b, rtol=1e-4, atol=1e-2), \
             (a - b).abs().max()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [(10000, 256, 602), (8192, 256, 256),
+                                   (999, 64, 100), (5000, 100, 70)])
+def test_dual_dgrad_gpu(shape):
+    """Fused MFMA dual-dgrad vs plain fp32 matmul reference."""
+    from pipegcn_amd import native
+
+    M, N, K = shape
+    torch.manual_seed(0)
+    g = torch.randn(M, N, device="cuda")
+    w1 = torch.randn(N, K, device="cuda")
+    w2 = torch.randn(N, K, device="cuda")
+    gx1, gx2 = native().dual_dgrad(g, w1, w2)
+    assert torch.allclose(gx1, g @ w1, rtol=1e-4, atol=1e-2), \
+        (gx1 - g @ w1).abs().max()
+    assert torch.allclose(gx2, g @ w2, rtol=1e-4, atol=1e-2)
