@@ -128,14 +128,17 @@ class _SageDualLinear(torch.autograd.Function):
         x1, x2, w1, w2 = ctx.saved_tensors
         g = g.contiguous()
         if g.is_cuda and g.dtype == torch.float32:
-            if g.size(1) >= 64:
+            if g.size(1) >= 64 and w1.size(1) <= 512:
                 # dgrad pair fused in one MFMA kernel (g tile staged once
-                # for both weight contractions, contiguous outputs)
+                # for both weight contractions; measured par with rocBLAS
+                # at K=256 — 117 TF both)
                 gx1, gx2 = native().dual_dgrad(g, w1.contiguous(),
                                                w2.contiguous())
             else:
-                # thin-N (e.g. the 41-class output layer): rocBLAS wins —
-                # ONE GEMM against [w1 ‖ w2] so g is still read once
+                # thin-N (the 41-class output layer) and wide-K (602:
+                # rocBLAS 112 vs our 95 TF — the unaligned 602-float rows
+                # and boundary k-tile cost us there): ONE GEMM against
+                # [w1 ‖ w2] so g is still read once
                 gx = g @ torch.cat((w1, w2), dim=1)
                 K = w1.size(1)
                 gx1, gx2 = gx[:, :K], gx[:, K:]
