@@ -174,3 +174,29 @@ def test_exact_vol_gain_verified_and_better():
     finally:
         os.environ.pop("PIPEGCN_PART_CHECK_VOL", None)
         os.environ.pop("PIPEGCN_PART_EXACT_VOL_MB", None)
+
+
+def test_partitioner_thread_count_invariance():
+    """The parallel-prefilter refinement and parallel coarsening must give
+    the SAME partition regardless of thread count (evaluation is a pure
+    function of the round snapshot; application is serial in node order)."""
+    import subprocess
+    import sys
+
+    code = """
+import torch
+torch.manual_seed(0)
+torch.set_num_threads({n})
+from pipegcn_amd.graph.partition import assign_partitions
+u = torch.randint(0, 4000, (30000,), generator=torch.Generator().manual_seed(1))
+v = torch.randint(0, 4000, (30000,), generator=torch.Generator().manual_seed(2))
+p = assign_partitions(u, v, 4000, 4, "metis", "vol", 3)
+print(hash(tuple(p.tolist())))
+"""
+    outs = []
+    for n in (1, 8):
+        r = subprocess.run([sys.executable, "-c", code.format(n=n)],
+                           capture_output=True, text=True)
+        assert r.returncode == 0, r.stderr
+        outs.append(r.stdout.strip())
+    assert outs[0] == outs[1], outs
