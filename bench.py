@@ -33,6 +33,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--shape", type=str, default="reddit",
                     help="synthetic shape (reddit/ogbn-products/yelp/small)")
+    ap.add_argument("--model", choices=["graphsage", "gcn"],
+                    default="graphsage")
     ap.add_argument("--n-hidden", type=int, default=256)
     ap.add_argument("--n-layers", type=int, default=3)
     ap.add_argument("--no-pipeline", action="store_true",
@@ -90,7 +92,8 @@ def main():
     from pipegcn_amd.graph.synthetic import SHAPES, synth_partition
     from pipegcn_amd.models.sage import GraphSAGE
     from pipegcn_amd.parallel import context as ctx
-    from pipegcn_amd.trainer import get_layer_size, precompute
+    from pipegcn_amd.trainer import (exchange_halo_values, get_layer_size,
+                                     precompute)
     from pipegcn_amd.utils.timer import comm_timer
     import torch.nn.functional as F
     import types
@@ -133,9 +136,19 @@ def main():
                            collect_stats=True, dtype=dtype,
                            solo_world=args.solo_of)
 
-    model = GraphSAGE(layer_size, F.relu, use_pp=args.use_pp, dropout=0.5,
-                      norm=norm, n_linear=0,
-                      train_size=part.n_train).to(device)
+    if args.model == "gcn":
+        from pipegcn_amd.models.gcn import GCN
+
+        if args.use_pp:
+            raise SystemExit("--use-pp supports graphsage only "
+                             "(reference parity)")
+        model = GCN(layer_size, F.relu, use_pp=False, dropout=0.5,
+                    norm=norm, n_linear=0,
+                    train_size=part.n_train).to(device)
+    else:
+        model = GraphSAGE(layer_size, F.relu, use_pp=args.use_pp,
+                          dropout=0.5, norm=norm, n_linear=0,
+                          train_size=part.n_train).to(device)
     if dtype != torch.float32:
         model = model.to(dtype)
     ctx.reducer.init(model)
@@ -157,6 +170,9 @@ def main():
         rp.ndata.pop("feat", None)
         phase("use-pp precompute done")
     in_deg = rp.ndata["in_degree"]
+    if args.model == "gcn":
+        # GCN's symmetric normalization needs halo-node degrees too
+        in_deg = exchange_halo_values(rp, in_deg)
     labels = rp.ndata["label"][: rp.num_train]
     model.train()
 
@@ -220,7 +236,9 @@ def main():
         baseline = 0.2660  # BASELINE.md Reddit epoch time (rank 0), other hw
         result = {
             "metric": (f"epoch time (s), {args.shape} "
-                       f"{args.n_layers}-layer GraphSAGE (full-graph, "
+                       f"{args.n_layers}-layer "
+                       f"{'GCN' if args.model == 'gcn' else 'GraphSAGE'}"
+                       " (full-graph, "
                        f"{'pipelined' if pipeline else 'vanilla'})"),
             "value": epoch_s,
             "unit": "s/epoch",
@@ -238,7 +256,7 @@ def main():
             "dtype": args.dtype,
             "data": "synthetic",
             "config": {
-                "model": f"graphsage-{args.n_layers}L-h{args.n_hidden}",
+                "model": f"{args.model}-{args.n_layers}L-h{args.n_hidden}",
                 "graph": args.shape,
                 "num_nodes": n,
                 "avg_degree": avg_deg,
