@@ -136,10 +136,11 @@ class _SageDualLinear(torch.autograd.Function):
             # once for both products; deterministic workspace reduce)
             gw1, gw2 = native().dual_wgrad(g, x1.contiguous(),
                                            x2.contiguous())
-            if g.size(1) >= 64 and w1.size(1) <= 512:
+            if g.size(1) >= 64 and w1.size(1) <= 256:
                 # dgrad pair fused in one MFMA kernel (g tile staged once
                 # for both weight contractions; measured par with rocBLAS
-                # at K=256 — 117 TF both)
+                # at K=256 — 117 TF both; rocBLAS wins at K=512 (142 vs
+                # 123 TF) and K=602 (112 vs 95), hence the cap)
                 gx1, gx2 = native().dual_dgrad(g, w1.contiguous(),
                                                w2.contiguous())
             else:
