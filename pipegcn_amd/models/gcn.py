@@ -31,21 +31,28 @@ class _SpmmSym(torch.autograd.Function):
     def forward(ctx_, graph: HaloGraph, feat, inv_sqrt_dst, inv_sqrt_all):
         ctx_.graph = graph
         ctx_.save_for_backward(inv_sqrt_dst, inv_sqrt_all)
-        # the source-side scale runs as ONE streaming row-multiply instead
-        # of a per-edge gather inside the SpMM (measured ~1 ms/call saved
-        # at reddit scale; see _SpmmMean.backward)
-        return ops.spmm(graph.csr,
-                        feat * inv_sqrt_all.unsqueeze(1).to(feat.dtype),
-                        inv_sqrt_dst)
+        # source-side scale: streaming row-multiply on dense graphs,
+        # fused per-edge gather on sparse-wide ones (cost model in
+        # _SpmmMean.backward)
+        if graph.csr.nnz > feat.numel():
+            return ops.spmm(graph.csr,
+                            feat * inv_sqrt_all.unsqueeze(1).to(feat.dtype),
+                            inv_sqrt_dst)
+        return ops.spmm(graph.csr, feat, inv_sqrt_dst,
+                        src_scale=inv_sqrt_all)
 
     @staticmethod
     def backward(ctx_, grad_out):
         inv_sqrt_dst, inv_sqrt_all = ctx_.saved_tensors
         g = ctx_.graph
-        grad_feat = ops.spmm(
-            g.csc,
-            grad_out * inv_sqrt_dst.unsqueeze(1).to(grad_out.dtype),
-            inv_sqrt_all)
+        if g.csc.nnz > grad_out.numel():
+            grad_feat = ops.spmm(
+                g.csc,
+                grad_out * inv_sqrt_dst.unsqueeze(1).to(grad_out.dtype),
+                inv_sqrt_all)
+        else:
+            grad_feat = ops.spmm(g.csc, grad_out.contiguous(),
+                                 inv_sqrt_all, src_scale=inv_sqrt_dst)
         return None, grad_feat, None, None
 
 

@@ -48,13 +48,19 @@ class _SpmmMean(torch.autograd.Function):
     def backward(ctx, grad_out: torch.Tensor):
         (inv_deg,) = ctx.saved_tensors
         g = ctx.graph
-        # D^{-1} applied as ONE streaming row-multiply on g (~0.06 ms at
-        # [233k,256]) instead of a per-EDGE src_scale gather inside the
-        # transpose SpMM (measured ~1 ms/call: fwd-no-scale 15.4 ms vs
-        # bwd-with-src-scale 16.3 ms on the same CSC shape)
-        grad_feat = spmm(g.csc,
-                         grad_out * inv_deg.unsqueeze(1).to(grad_out.dtype),
-                         None)
+        # D^{-1} either as ONE streaming row-multiply on g or as the
+        # fused per-EDGE src_scale gather: the multiply costs ~3 passes
+        # over [rows,F], the gather ~one 4B load per edge plus its
+        # latency slot — dense graphs (reddit: nnz 115M > numel 60M,
+        # measured +1 ms/call for pre-scale) want the multiply, sparse
+        # wide ones (yelp: nnz 14M << numel 367M) want the gather
+        if g.csc.nnz > grad_out.numel():
+            grad_feat = spmm(
+                g.csc,
+                grad_out * inv_deg.unsqueeze(1).to(grad_out.dtype), None)
+        else:
+            grad_feat = spmm(g.csc, grad_out.contiguous(), None,
+                             src_scale=inv_deg)
         return None, grad_feat, None
 
 
