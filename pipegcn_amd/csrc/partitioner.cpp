@@ -34,6 +34,8 @@
 
 #include <ATen/Parallel.h>
 
+#include <atomic>
+
 #include <algorithm>
 #include <chrono>
 #include <cstdio>
@@ -79,29 +81,112 @@ struct Graph {
   int64_t nw(int64_t x) const { return nw_s.empty() ? 1 : nw_s[x]; }
 };
 
+// Parallel SUITOR matching (Manne-Bisseling) with hash-de-tied weights.
+// Round-1 finding: on unit-weight graphs every proposal ties, each tie
+// displaces a prior suitor, and every displaced node rescans its full
+// neighbor list — a 20x slowdown at level 0. The fix (designed then,
+// implemented now): a strict TOTAL order on edge keys,
+//    key = (min(ew,0xFFFF)<<16 | hash16(u,v)) << 32 | proposer,
+// so first proposals almost always stick and the final matching is the
+// unique greedy matching of the de-tied weights — deterministic across
+// thread counts and schedules by construction. Leftover unmatched nodes
+// fall through to the same 2-hop pass as the serial path.
+int64_t suitor_matching(const Graph& g, std::vector<int32_t>& match) {
+  const int64_t n = g.n;
+  std::vector<std::atomic<uint64_t>> suitor(n);
+  at::parallel_for(0, n, 1 << 16, [&](int64_t b, int64_t en) {
+    for (int64_t i = b; i < en; ++i)
+      suitor[i].store(0, std::memory_order_relaxed);
+  });
+  auto edge_key = [&](int64_t u, int64_t v, int64_t ew) -> uint64_t {
+    const uint64_t a = u < v ? u : v;
+    const uint64_t bb = u < v ? v : u;
+    uint64_t h = (a + 0x9E3779B97F4A7C15ull) * 0xBF58476D1CE4E5B9ull;
+    h ^= (bb + 0x94D049BB133111EBull) * 0x2545F4914F6CDD1Dull;
+    h ^= h >> 29;
+    const uint64_t w = (uint64_t)std::min<int64_t>(ew, 0xFFFF);
+    const uint64_t wkey = (w << 16) | (h & 0xFFFF);
+    // +1 on the proposer so key 0 means "no suitor yet"
+    return (wkey << 32) | (uint64_t)(u + 1);
+  };
+  at::parallel_for(0, n, 4096, [&](int64_t b, int64_t en) {
+    for (int64_t start = b; start < en; ++start) {
+      int64_t x = start;
+      while (x >= 0) {
+        // best eligible neighbor of x: highest key that still beats
+        // v's current suitor
+        int64_t best_v = -1;
+        uint64_t best_key = 0;
+        for (int64_t e = g.indptr[x]; e < g.indptr[x + 1]; ++e) {
+          const int32_t v = g.indices[e];
+          if (v == x) continue;
+          const uint64_t k = edge_key(x, v, g.ew(e));
+          if (k <= best_key) continue;
+          if (k > suitor[v].load(std::memory_order_relaxed)) {
+            best_key = k;
+            best_v = v;
+          }
+        }
+        if (best_v < 0) break;
+        uint64_t old = suitor[best_v].load(std::memory_order_relaxed);
+        if (old >= best_key) continue;  // raced: re-scan x
+        if (suitor[best_v].compare_exchange_weak(
+                old, best_key, std::memory_order_acq_rel)) {
+          // displaced proposer (if any) must re-propose elsewhere
+          x = (int64_t)(old & 0xFFFFFFFFull) - 1;
+        }
+        // CAS failure: loop re-scans x against the fresher state
+      }
+    }
+  });
+  // mutual suitors are matched; everyone else stays -1 for the 2-hop pass
+  match.assign(n, -1);
+  at::parallel_for(0, n, 1 << 16, [&](int64_t b, int64_t en) {
+    for (int64_t u = b; u < en; ++u) {
+      const uint64_t su = suitor[u].load(std::memory_order_relaxed);
+      const int64_t p = (int64_t)(su & 0xFFFFFFFFull) - 1;
+      if (p < 0) continue;
+      const uint64_t sp = suitor[p].load(std::memory_order_relaxed);
+      if ((int64_t)(sp & 0xFFFFFFFFull) - 1 == u)
+        match[u] = (int32_t)p;  // mutual (written from both sides)
+    }
+  });
+  int64_t matched = 0;
+  for (int64_t u = 0; u < n; ++u)
+    if (match[u] >= 0) matched++;
+  return matched;
+}
+
 // heavy-edge matching; returns coarse count and fine->coarse map
 int64_t heavy_edge_matching(const Graph& g, std::mt19937_64& rng,
                             std::vector<int32_t>& cmap) {
   const int64_t n = g.n;
-  std::vector<int32_t> match(n, -1);
-  std::vector<int32_t> order(n);
-  std::iota(order.begin(), order.end(), 0);
-  std::shuffle(order.begin(), order.end(), rng);
-  for (int32_t u : order) {
-    if (match[u] >= 0) continue;
-    int32_t best = -1;
-    int64_t best_w = -1;
-    for (int64_t e = g.indptr[u]; e < g.indptr[u + 1]; ++e) {
-      const int32_t v = g.indices[e];
-      if (v == u || match[v] >= 0) continue;
-      if (g.ew(e) > best_w) {
-        best_w = g.ew(e);
-        best = v;
+  std::vector<int32_t> match;
+  const int64_t suitor_min = env_int("PIPEGCN_PART_SUITOR_MIN", 2000000);
+  if (n >= suitor_min) {
+    // big levels: parallel Suitor with hash-de-tied keys (see above)
+    suitor_matching(g, match);
+  } else {
+    match.assign(n, -1);
+    std::vector<int32_t> order(n);
+    std::iota(order.begin(), order.end(), 0);
+    std::shuffle(order.begin(), order.end(), rng);
+    for (int32_t u : order) {
+      if (match[u] >= 0) continue;
+      int32_t best = -1;
+      int64_t best_w = -1;
+      for (int64_t e = g.indptr[u]; e < g.indptr[u + 1]; ++e) {
+        const int32_t v = g.indices[e];
+        if (v == u || match[v] >= 0) continue;
+        if (g.ew(e) > best_w) {
+          best_w = g.ew(e);
+          best = v;
+        }
       }
-    }
-    if (best >= 0) {
-      match[u] = best;
-      match[best] = u;
+      if (best >= 0) {
+        match[u] = best;
+        match[best] = u;
+      }
     }
   }
   // 2-hop pass (METIS-style): heavy-tailed graphs stall the 1-hop pass —

@@ -200,3 +200,31 @@ print(hash(tuple(p.tolist())))
         assert r.returncode == 0, r.stderr
         outs.append(r.stdout.strip())
     assert outs[0] == outs[1], outs
+
+
+def test_suitor_matching_deterministic_and_sane():
+    """Force the parallel Suitor matching path (PIPEGCN_PART_SUITOR_MIN)
+    and check thread-count determinism + partition sanity on it."""
+    import subprocess
+    import sys
+
+    code = """
+import os, torch
+os.environ["PIPEGCN_PART_SUITOR_MIN"] = "1000"  # force Suitor
+torch.set_num_threads({n})
+from pipegcn_amd.graph.partition import assign_partitions
+g1 = torch.Generator().manual_seed(11)
+u = torch.randint(0, 20000, (160000,), generator=g1)
+v = torch.randint(0, 20000, (160000,), generator=g1)
+p = assign_partitions(u, v, 20000, 4, "metis", "vol", 3)
+sizes = torch.bincount(p.long(), minlength=4)
+assert sizes.max() <= int(20000 / 4 * 1.05) + 1, sizes
+print(hash(tuple(p.tolist())))
+"""
+    outs = []
+    for n in (1, 8):
+        r = subprocess.run([sys.executable, "-c", code.format(n=n)],
+                           capture_output=True, text=True)
+        assert r.returncode == 0, r.stderr
+        outs.append(r.stdout.strip())
+    assert outs[0] == outs[1], outs
