@@ -101,11 +101,15 @@ __global__ void dropout_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
         *reinterpret_cast<float4*>(reinterpret_cast<float*>(y) + i0 + 4) =
             ob;
       } else {
+        // bf16: the 8 elements are ONE 16B transaction (i0 = 8t)
+        const uint4 a = *reinterpret_cast<const uint4*>(x + i0);
+        const T* ap = reinterpret_cast<const T*>(&a);
+        uint4 o;
+        T* op = reinterpret_cast<T*>(&o);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const float v = (m >> j) & 1 ? to_f32(x[i0 + j]) * scale : 0.f;
-          from_f32(v, &y[i0 + j]);
-        }
+        for (int j = 0; j < 8; ++j)
+          from_f32((m >> j) & 1 ? to_f32(ap[j]) * scale : 0.f, &op[j]);
+        *reinterpret_cast<uint4*>(y + i0) = o;
       }
     } else {
       for (int j = 0; i0 + j < n; ++j) {
@@ -146,11 +150,14 @@ __global__ void dropout_bwd_kernel(const T* __restrict__ dy,
         *reinterpret_cast<float4*>(reinterpret_cast<float*>(dx) + i0 + 4) =
             ob;
       } else {
+        const uint4 a = *reinterpret_cast<const uint4*>(dy + i0);
+        const T* ap = reinterpret_cast<const T*>(&a);
+        uint4 o;
+        T* op = reinterpret_cast<T*>(&o);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const float v = (m >> j) & 1 ? to_f32(dy[i0 + j]) * scale : 0.f;
-          from_f32(v, &dx[i0 + j]);
-        }
+        for (int j = 0; j < 8; ++j)
+          from_f32((m >> j) & 1 ? to_f32(ap[j]) * scale : 0.f, &op[j]);
+        *reinterpret_cast<uint4*>(dx + i0) = o;
       }
     } else {
       for (int j = 0; i0 + j < n; ++j) {
