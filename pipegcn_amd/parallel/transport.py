@@ -128,6 +128,12 @@ def exchange_index_lists(wanted: List[Optional[torch.Tensor]],
     out: List[Optional[torch.Tensor]] = [None] * size
     use_cuda = dist.get_backend(group) == "nccl"
     dev = "cuda" if use_cuda else "cpu"
+    if use_cuda:
+        # communicator warmup: this is the FIRST op on the group in the
+        # setup path — initialize the RCCL communicator from a plain
+        # collective rather than from a grouped P2P (the robust pattern;
+        # mirrors Buffer.init_buffer)
+        dist.all_reduce(torch.zeros(1, device=dev), group=group)
     # phase 1: sizes — ONE batched group over all ring peers (unmatched
     # singleton isend/recv is a hang-prone pattern on NCCL/RCCL; grouped
     # P2P is what the data path uses too)
