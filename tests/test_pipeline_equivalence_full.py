@@ -279,3 +279,14 @@ def test_gcn_pipelined_matches_simulator(tmp_path):
     """GCN (sym-normalized SpMM + halo-degree exchange) under the same
     pipelined staleness spec."""
     _check(tmp_path, GCN_CFG)
+
+
+NLINEAR_CFG = dict(enable_pipeline=True, n_layers=4, n_linear=2,
+                   n_hidden=16, norm="layer", n_epochs=5, lr=0.05)
+
+
+def test_linear_tail_matches_simulator(tmp_path):
+    """Yelp-style layer stack (2 conv + --n-linear 2 tail layers,
+    scripts/yelp.sh / reference scripts) under the pipelined staleness
+    spec — the buffer must cover ONLY the conv layers."""
+    _check(tmp_path, NLINEAR_CFG)
