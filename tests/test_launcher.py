@@ -72,6 +72,17 @@ def test_main_yelp_style_config(tmp_path):
     assert "Accuracy" in out  # micro-F1 eval path
 
 
+def test_main_reddit_recipe(tmp_path):
+    """The reference's headline reddit.sh recipe flag-for-flag (4 layers,
+    hidden, dropout 0.5, --inductive --enable-pipeline --use-pp) at
+    synth-small scale (scripts/reddit.sh; /root/reference/scripts/reddit.sh)."""
+    out = run_main(tmp_path, ["--n-layers", "4", "--dropout", "0.5",
+                              "--lr", "0.01", "--inductive",
+                              "--enable-pipeline", "--use-pp"])
+    assert "Epoch" in out
+    assert "Accuracy" in out
+
+
 def test_multi_node_launcher(tmp_path):
     """Simulate 2 nodes on localhost: each runs main.py with its own
     --node-rank and --parts-per-node 1; rendezvous over MASTER_ADDR/PORT
