@@ -21,8 +21,10 @@
 //     symmetrized input duplicates every edge; duplicates would also
 //     break the distinct-neighbor volume logic) and carries unit edge/
 //     node weights implicitly (no 2x int32/int64 arrays at full scale);
-//   - coarsening is a deterministic parallel two-pass; refinement and
-//     matching are serial sweeps (measured in profiles/).
+//   - coarsening is a deterministic parallel two-pass; matching on big
+//     levels is parallel Suitor with hash-de-tied keys; refinement is a
+//     parallel-prefilter boundary queue with serial live application
+//     (all thread-count-deterministic; measured in profiles/).
 //
 // PIPEGCN_PART_CHECK_VOL=1: verify every accepted vol move against a
 // brute-force local volume recomputation (the exact-gain correctness
