@@ -63,9 +63,13 @@ def main():
     if args.solo_of > 1:
         # sizing runs push toward the 288 GB HBM ceiling where caching-
         # allocator fragmentation (tens of GB reserved-but-unallocated)
-        # turns a fitting run into an OOM; expandable segments remove it
-        os.environ.setdefault("PYTORCH_ALLOC_CONF",
-                              "expandable_segments:True")
+        # turns a fitting run into an OOM. expandable_segments is not
+        # supported by this ROCm build; cap large-block splitting and
+        # garbage-collect the cache under pressure instead (both
+        # supported by the HIP caching allocator)
+        os.environ.setdefault(
+            "PYTORCH_ALLOC_CONF",
+            "max_split_size_mb:512,garbage_collection_threshold:0.8")
 
     env_world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -185,6 +189,12 @@ def main():
         ctx.reducer.synchronize(part.n_train)
         optimizer.step()
         comm_timer.clear()
+        if solo:
+            # sizing mode runs within a few % of the memory ceiling:
+            # return the freed large blocks to HIP each epoch so
+            # fragmentation cannot accumulate (costs ~ms; sizing runs
+            # measure FIT, the timed configs never take this branch)
+            torch.cuda.empty_cache()
         return loss
 
     def barrier_sync():
@@ -214,6 +224,8 @@ def main():
         optimizer.step()
         wait_s += comm_timer.tot_time()
         comm_timer.clear()
+        if solo:
+            torch.cuda.empty_cache()  # see step(): sizing-mode only
     # drain the pipelined buffer queue so the final epoch's boundary
     # transfers are inside the timed region (they belong to the epoch)
     ctx.buffer.synchronize()
