@@ -72,7 +72,15 @@ def spmm(csr: CSR, feat: torch.Tensor,
             csr._lpt_choice = cache
         use_lpt = cache.get(feat.shape[1])
         if use_lpt is None:
-            use_lpt = _tune_row_order(csr, feat, s, ss)
+            # the A/B makes 6 transient outputs; near the memory ceiling
+            # (papers100M sizing: tuning fires mid-backward at peak) that
+            # transient OOMs or inflates peak — keep the LPT default there
+            free, _ = torch.cuda.mem_get_info(feat.device)
+            out_bytes = csr.num_rows * feat.shape[1] * feat.element_size()
+            if free < 3 * out_bytes + (2 << 30):
+                use_lpt = True
+            else:
+                use_lpt = _tune_row_order(csr, feat, s, ss)
             cache[feat.shape[1]] = use_lpt
         if not use_lpt:
             ro = torch.Tensor()
