@@ -378,3 +378,39 @@ def test_dual_dgrad_gpu(shape):
     assert torch.allclose(gx1, g @ w1, rtol=1e-4, atol=1e-2), \
         (gx1 - g @ w1).abs().max()
     assert torch.allclose(gx2, g @ w2, rtol=1e-4, atol=1e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("f", [33, 602, 1000, 1024])
+def test_layer_norm_relu_odd_widths_gpu(f):
+    """LN kernel edge sizes: odd F, F not a multiple of 64*VEC, the
+    F<=1024 register cap boundary."""
+    torch.manual_seed(f)
+    x = torch.randn(777, f, device="cuda", requires_grad=True)
+    ln = torch.nn.LayerNorm(f).cuda()
+    out = ops.layer_norm_relu(x, ln, relu=True)
+    ref = torch.nn.functional.relu(
+        torch.nn.functional.layer_norm(x, (f,), ln.weight, ln.bias, ln.eps))
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+    g = torch.randn_like(out)
+    out.backward(g)
+    gx = x.grad.clone()
+    x.grad = None
+    ln.weight.grad = ln.bias.grad = None
+    torch.nn.functional.relu(
+        torch.nn.functional.layer_norm(x, (f,), ln.weight, ln.bias,
+                                       ln.eps)).backward(g)
+    assert torch.allclose(gx, x.grad, atol=1e-3, rtol=1e-3)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("p", [0.01, 0.9])
+def test_fused_dropout_extreme_p_gpu(p):
+    torch.manual_seed(0)
+    x = torch.randn(100001, device="cuda")  # odd length: tail path
+    from pipegcn_amd import native
+    y, mask = native().dropout_fwd(x, p, 7)
+    frac = (y != 0).float().mean().item()
+    assert abs(frac - (1 - p)) < 0.02, frac
+    dx = native().dropout_bwd(y, mask, p)
+    assert dx.shape == x.shape
