@@ -87,3 +87,28 @@ def test_partitioned_aggregation_matches_global(tmp_path_factory, cfg):
         got = out[st_["new_id"]]
         assert torch.allclose(got, ref[offs[i]: offs[i + 1]], atol=1e-4), \
             (i, (got - ref[offs[i]: offs[i + 1]]).abs().max())
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(min_value=0, max_value=2 ** 31),
+       st.integers(min_value=2, max_value=6))
+def test_exact_vol_oracle_fuzz(seed, nparts):
+    """Every accepted vol-refinement move on arbitrary random graphs is
+    verified in-C++ against a brute-force local volume recomputation
+    (TORCH_CHECK aborts on any gain mismatch)."""
+    import os
+
+    from pipegcn_amd.graph.partition import assign_partitions
+
+    g = torch.Generator().manual_seed(seed)
+    n = 500 + seed % 1500
+    e = 4 * n
+    u = torch.randint(0, n, (e,), generator=g)
+    v = torch.randint(0, n, (e,), generator=g)
+    os.environ["PIPEGCN_PART_CHECK_VOL"] = "1"
+    try:
+        p = assign_partitions(u, v, n, nparts, "metis", "vol", seed % 100)
+    finally:
+        os.environ.pop("PIPEGCN_PART_CHECK_VOL", None)
+    sizes = torch.bincount(p.long(), minlength=nparts)
+    assert sizes.max() <= int(n / nparts * 1.05) + 1
